@@ -1,0 +1,168 @@
+"""Python dispatch layer over the MI355X native kernel pack.
+
+Every op has two paths:
+- device tensors -> the hand-written CDNA4 HIP kernels in `mi355x_ddp._C`
+  (MFMA-tiled linear, CE/MSE, fused SGD, bucket copies). If the extension
+  is missing on a GPU host this layer raises — there is NO silent eager
+  fallback on the GPU.
+- CPU tensors -> plain PyTorch reference implementations, used by the
+  CPU-only test tier (gloo, world_size>1) and the single_gpu CPU plumbing
+  config (BASELINE.json config 1).
+
+The CPU implementations double as the numerics references the GPU kernels
+are tested against (tests/test_kernels_gpu.py).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+_EXT = None
+_EXT_ERR: Optional[str] = None
+
+
+def _load_ext():
+    global _EXT, _EXT_ERR
+    if _EXT is None and _EXT_ERR is None:
+        try:
+            from mi355x_ddp import _C  # built in-tree by setup.py build_ext --inplace
+            _EXT = _C
+        except ImportError as e:  # pragma: no cover
+            _EXT_ERR = str(e)
+    return _EXT
+
+
+def ext():
+    """The native extension; raises loudly if unavailable."""
+    m = _load_ext()
+    if m is None:
+        raise RuntimeError(
+            "mi355x_ddp._C native extension is not built — run "
+            "`python setup.py build_ext --inplace` (hipcc, gfx950). "
+            f"Import error: {_EXT_ERR}")
+    return m
+
+
+def has_ext() -> bool:
+    return _load_ext() is not None
+
+
+# ---------------------------------------------------------------------------
+# Linear (reference model: torch.nn.Linear(20,1), single_gpu.py:50)
+# ---------------------------------------------------------------------------
+class _HipLinearFn(torch.autograd.Function):
+    """y = x @ w^T + b with hand-written MFMA kernels."""
+
+    @staticmethod
+    def forward(ctx, x, w, b):
+        ctx.save_for_backward(x, w)
+        ctx.has_bias = b is not None
+        return ext().linear_fwd(x.contiguous(), w, b)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, w = ctx.saved_tensors
+        dy = dy.contiguous()
+        dx = dw = db = None
+        if ctx.needs_input_grad[0]:
+            dx = ext().linear_bwd_input(dy, w)
+        if ctx.needs_input_grad[1] or (ctx.has_bias and ctx.needs_input_grad[2]):
+            dw = torch.empty_like(w)
+            db = torch.empty(w.shape[0], dtype=w.dtype, device=w.device)
+            ext().linear_bwd_weight(x, dy, dw, db, False)
+        if not ctx.has_bias:
+            db = None
+        return dx, dw, db
+
+
+def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None):
+    if x.is_cuda:
+        return _HipLinearFn.apply(x, w, b)
+    return F.linear(x, w, b)
+
+
+# ---------------------------------------------------------------------------
+# Losses (reference: CE at single_gpu.py:24, MSE at multinode_torchrun.py:46;
+# both take float probability/value targets of the same shape as the output)
+# ---------------------------------------------------------------------------
+class _HipCEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, t):
+        loss, probs, tsum = ext().ce_fwd(y, t)
+        ctx.save_for_backward(probs, t, tsum)
+        return loss
+
+    @staticmethod
+    def backward(ctx, gout):
+        probs, t, tsum = ctx.saved_tensors
+        # keep capture-safe: no D2H sync on gout — scale on device instead
+        dy = ext().ce_bwd(probs, t, tsum, 1.0)
+        if torch.is_tensor(gout):
+            dy = dy * gout
+        else:
+            dy = dy * float(gout)
+        return dy, None
+
+
+def cross_entropy(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    """torch.nn.CrossEntropyLoss()(output, target) with probability targets.
+
+    Note the reference's toy case is degenerate (C=1 -> loss == 0, grads
+    == 0; SURVEY §2.1 'Degenerate loss') — both paths reproduce torch's
+    exact semantics for it.
+    """
+    if output.is_cuda:
+        return _HipCEFn.apply(output, target.to(output.dtype))
+    return F.cross_entropy(output, target.to(output.dtype))
+
+
+class _HipMSEFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, y, t):
+        ctx.save_for_backward(y, t)
+        return ext().mse_fwd(y, t)
+
+    @staticmethod
+    def backward(ctx, gout):
+        y, t = ctx.saved_tensors
+        dy = ext().mse_bwd(y, t, 1.0)
+        if torch.is_tensor(gout):
+            dy = dy * gout
+        else:
+            dy = dy * float(gout)
+        return dy, None
+
+
+def mse_loss(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
+    if output.is_cuda:
+        return _HipMSEFn.apply(output, target.to(output.dtype))
+    return F.mse_loss(output, target.to(output.dtype))
+
+
+# ---------------------------------------------------------------------------
+# Flat-bucket primitives used by the reducer / fused optimizer
+# ---------------------------------------------------------------------------
+def sgd_flat_(param_flat: torch.Tensor, grad_flat: torch.Tensor, lr: float,
+              zero_grad: bool = True) -> None:
+    """p -= lr*g over a flat bucket, grad zeroing folded in (SURVEY N8+N9)."""
+    if param_flat.is_cuda:
+        ext().sgd_flat(param_flat, grad_flat, lr, zero_grad)
+    else:
+        param_flat.add_(grad_flat, alpha=-lr)
+        if zero_grad:
+            grad_flat.zero_()
+
+
+def cpu_linear_bwd_weight(x, dy, dw, db, accumulate=False):
+    """CPU reference for the bwd-weight kernel (used in numerics tests)."""
+    w_new = dy.t() @ x
+    b_new = dy.sum(dim=0)
+    if accumulate:
+        dw += w_new
+        db += b_new
+    else:
+        dw.copy_(w_new)
+        db.copy_(b_new)

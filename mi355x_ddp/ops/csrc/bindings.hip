@@ -1,0 +1,52 @@
+// Python bindings for the mi355x_ddp native layer (kernels + RCCL comm).
+#include <torch/extension.h>
+
+#include "ops.h"
+#include "rccl_comm.h"
+
+namespace py = pybind11;
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X-native kernels and RCCL communicator for mi355x_ddp";
+
+  m.def("linear_fwd", &mi355x::linear_fwd, py::arg("x"), py::arg("w"),
+        py::arg("bias") = c10::nullopt);
+  m.def("linear_bwd_weight", &mi355x::linear_bwd_weight, py::arg("x"),
+        py::arg("dy"), py::arg("dw"), py::arg("db"),
+        py::arg("accumulate") = false);
+  m.def("linear_bwd_input", &mi355x::linear_bwd_input);
+  m.def("ce_fwd", &mi355x::ce_fwd);
+  m.def("ce_bwd", &mi355x::ce_bwd);
+  m.def("mse_fwd", &mi355x::mse_fwd);
+  m.def("mse_bwd", &mi355x::mse_bwd);
+  m.def("sgd_flat", &mi355x::sgd_flat, py::arg("param_flat"),
+        py::arg("grad_flat"), py::arg("lr"), py::arg("zero_grad") = true);
+  m.def("build_copy_plan",
+        [](const std::vector<torch::Tensor>& tensors,
+           const std::vector<int64_t>& offsets, int device) {
+          return mi355x::build_copy_plan(tensors, offsets,
+                                         torch::Device(torch::kCUDA, device));
+        });
+  m.def("flatten_into", &mi355x::flatten_into, py::arg("bucket"),
+        py::arg("plan"), py::arg("total_blocks"), py::arg("zero_src") = false);
+  m.def("unflatten_from", &mi355x::unflatten_from);
+  m.def("toy_fused_fwd_bwd", &mi355x::toy_fused_fwd_bwd, py::arg("x"),
+        py::arg("t"), py::arg("param_flat"), py::arg("grad_flat"),
+        py::arg("loss_out"), py::arg("use_mse") = true);
+
+  py::class_<mi355x::RcclComm>(m, "RcclComm")
+      .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),
+           py::arg("rank"), py::arg("world"), py::arg("device"),
+           py::call_guard<py::gil_scoped_release>())
+      .def_static("make_unique_id", [] {
+        return py::bytes(mi355x::RcclComm::make_unique_id());
+      })
+      .def_property_readonly("rank", &mi355x::RcclComm::rank)
+      .def_property_readonly("world", &mi355x::RcclComm::world)
+      .def("all_reduce_avg", &mi355x::RcclComm::all_reduce_avg)
+      .def("all_reduce_avg_inline", &mi355x::RcclComm::all_reduce_avg_inline)
+      .def("broadcast", &mi355x::RcclComm::broadcast)
+      .def("join_compute", &mi355x::RcclComm::join_compute)
+      .def("barrier", &mi355x::RcclComm::barrier,
+           py::call_guard<py::gil_scoped_release>());
+}

@@ -1,0 +1,553 @@
+// MI355X (gfx950, CDNA4) kernel pack for mi355x_ddp — written directly in
+// HIP for CDNA4 (64-wide wavefronts, MFMA via __builtin_amdgcn_mfma_*,
+// LDS staging). No CUDA compatibility paths, no hipify.
+//
+// Re-implements the native components the reference pulls in implicitly
+// (SURVEY.md §2.2): N6 GEMM fwd/bwd for the toy Linear (reference
+// single_gpu.py:23,25 / nn.Linear(20,1) at :50), N7 CE/MSE loss kernels
+// (single_gpu.py:24, multinode_torchrun.py:46), N8 fused SGD
+// (single_gpu.py:26), N9 zero_grad folded into the bucket lifecycle
+// (single_gpu.py:22), and the reducer's bucket flatten/unflatten (N3).
+//
+// f32 matmuls use the exact-f32 MFMA `v_mfma_f32_16x16x4_f32`
+// (f32-in/f32-acc at the f32 vector rate): the toy shapes are single-wave
+// and latency-bound, so the MFMA path buys kernel-count and issue-slot
+// economy, not FLOPs — see SURVEY.md §7 step 2.
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#include "ops.h"
+
+#define HIP_OK(expr)                                                          \
+  do {                                                                        \
+    hipError_t _e = (expr);                                                   \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));      \
+  } while (0)
+
+namespace mi355x {
+
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+static inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+static inline int cdiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
+
+// ---------------------------------------------------------------------------
+// Linear forward: Y[B,N] = X[B,K] @ W[N,K]^T + bias[N]
+// One wave (64 lanes) per 16x16 output tile; K-loop of mfma_f32_16x16x4f32.
+// Lane maps (cdna_hip_programming.md §3): A[l&15][k=l>>4], B[k=l>>4][l&15],
+// C/D col=lane&15, row=(lane>>4)*4+reg.
+// ---------------------------------------------------------------------------
+__global__ void k_linear_fwd_f32(const float* __restrict__ X,
+                                 const float* __restrict__ W,
+                                 const float* __restrict__ bias,
+                                 float* __restrict__ Y,
+                                 int B, int K, int N) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;  // 4 waves/block, each one M-tile
+  const int tile_m = blockIdx.x * 4 + wave;
+  const int tile_n = blockIdx.y;
+  if (tile_m * 16 >= B) return;
+  const int r = lane & 15;
+  const int q = lane >> 4;
+  const int m = tile_m * 16 + r;   // A row for this lane
+  const int n = tile_n * 16 + r;   // B column (j) for this lane
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k0 = 0; k0 < K; k0 += 4) {
+    const int k = k0 + q;
+    const float a = (m < B && k < K) ? X[(size_t)m * K + k] : 0.f;
+    const float b = (n < N && k < K) ? W[(size_t)n * K + k] : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+  const int col = tile_n * 16 + r;
+  if (col < N) {
+    const float bv = bias ? bias[col] : 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = tile_m * 16 + q * 4 + i;
+      if (row < B) Y[(size_t)row * N + col] = acc[i] + bv;
+    }
+  }
+}
+
+torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat, "x must be f32 on device");
+  TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
+              "shape mismatch for linear");
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const int B = (int)xc.size(0), K = (int)xc.size(1), N = (int)wc.size(0);
+  auto y = at::empty({B, N}, x.options());
+  const float* bp = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value()) { bc = bias->contiguous(); bp = bc.data_ptr<float>(); }
+  dim3 grid(cdiv(B, 64), cdiv(N, 16));
+  hipLaunchKernelGGL(k_linear_fwd_f32, grid, dim3(256), 0, cur_stream(),
+                     xc.data_ptr<float>(), wc.data_ptr<float>(), bp,
+                     y.data_ptr<float>(), B, K, N);
+  HIP_OK(hipGetLastError());
+  return y;
+}
+
+// ---------------------------------------------------------------------------
+// Linear weight/bias backward: dW[N,K] = dY[B,N]^T @ X[B,K]; db[N] = sum_i dY.
+// Output tile (16 rows of N) x (16 cols of K) per wave; contraction over B.
+// A[j][i] = dY[i][j]; B-operand[i][k] = X[i][k].
+// ---------------------------------------------------------------------------
+__global__ void k_linear_bwd_w_f32(const float* __restrict__ X,
+                                   const float* __restrict__ dY,
+                                   float* __restrict__ dW,
+                                   float* __restrict__ dB,
+                                   int B, int K, int N, int accumulate) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tile_k = blockIdx.x * 4 + wave;  // K-tile (output cols)
+  const int tile_n = blockIdx.y;             // N-tile (output rows)
+  const int r = lane & 15;
+  const int q = lane >> 4;
+  const int j = tile_n * 16 + r;  // dY column for A operand
+  const int k = tile_k * 16 + r;  // X column for B operand
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  float bsum = 0.f;  // partial db over this lane's i-chunk
+  if (tile_k * 16 < K) {
+    for (int i0 = 0; i0 < B; i0 += 4) {
+      const int i = i0 + q;
+      const float a = (i < B && j < N) ? dY[(size_t)i * N + j] : 0.f;
+      const float b = (i < B && k < K) ? X[(size_t)i * K + k] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+      if (tile_k == 0) bsum += a;  // reuse the loaded dY for db
+    }
+    const int col = tile_k * 16 + r;
+    if (col < K) {
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = tile_n * 16 + q * 4 + i;
+        if (row < N) {
+          const size_t off = (size_t)row * K + col;
+          dW[off] = acc[i] + (accumulate ? dW[off] : 0.f);
+        }
+      }
+    }
+  } else if (tile_k == 0) {
+    for (int i0 = 0; i0 < B; i0 += 4) {
+      const int i = i0 + q;
+      bsum += (i < B && j < N) ? dY[(size_t)i * N + j] : 0.f;
+    }
+  }
+  if (tile_k == 0 && dB) {
+    // combine the four q-chunks of lane-row r: lanes r, r+16, r+32, r+48
+    bsum += __shfl_xor(bsum, 16, 64);
+    bsum += __shfl_xor(bsum, 32, 64);
+    if (q == 0 && j < N) {
+      if (accumulate) dB[j] += bsum; else dB[j] = bsum;
+    }
+  }
+}
+
+void linear_bwd_weight(torch::Tensor x, torch::Tensor dy,
+                       torch::Tensor dw, torch::Tensor db, bool accumulate) {
+  auto xc = x.contiguous();
+  auto dyc = dy.contiguous();
+  TORCH_CHECK(dw.is_contiguous() && db.is_contiguous(), "grad views must be contiguous");
+  const int B = (int)xc.size(0), K = (int)xc.size(1), N = (int)dyc.size(1);
+  dim3 grid(cdiv(K, 64) , cdiv(N, 16));
+  hipLaunchKernelGGL(k_linear_bwd_w_f32, grid, dim3(256), 0, cur_stream(),
+                     xc.data_ptr<float>(), dyc.data_ptr<float>(),
+                     dw.data_ptr<float>(), db.numel() ? db.data_ptr<float>() : nullptr,
+                     B, K, N, accumulate ? 1 : 0);
+  HIP_OK(hipGetLastError());
+}
+
+// dX[B,K] = dY[B,N] @ W[N,K]
+__global__ void k_linear_bwd_x_f32(const float* __restrict__ dY,
+                                   const float* __restrict__ W,
+                                   float* __restrict__ dX,
+                                   int B, int K, int N) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tile_m = blockIdx.x * 4 + wave;
+  const int tile_k = blockIdx.y;
+  if (tile_m * 16 >= B) return;
+  const int r = lane & 15;
+  const int q = lane >> 4;
+  const int m = tile_m * 16 + r;   // dY row
+  const int kc = tile_k * 16 + r;  // W column for B operand
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int j0 = 0; j0 < N; j0 += 4) {
+    const int j = j0 + q;
+    const float a = (m < B && j < N) ? dY[(size_t)m * N + j] : 0.f;
+    const float b = (j < N && kc < K) ? W[(size_t)j * K + kc] : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+  }
+  const int col = tile_k * 16 + r;
+  if (col < K) {
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = tile_m * 16 + q * 4 + i;
+      if (row < B) dX[(size_t)row * K + col] = acc[i];
+    }
+  }
+}
+
+torch::Tensor linear_bwd_input(torch::Tensor dy, torch::Tensor w) {
+  auto dyc = dy.contiguous();
+  auto wc = w.contiguous();
+  const int B = (int)dyc.size(0), N = (int)dyc.size(1), K = (int)wc.size(1);
+  auto dx = at::empty({B, K}, dy.options());
+  dim3 grid(cdiv(B, 64), cdiv(K, 16));
+  hipLaunchKernelGGL(k_linear_bwd_x_f32, grid, dim3(256), 0, cur_stream(),
+                     dyc.data_ptr<float>(), wc.data_ptr<float>(),
+                     dx.data_ptr<float>(), B, K, N);
+  HIP_OK(hipGetLastError());
+  return dx;
+}
+
+// ---------------------------------------------------------------------------
+// Cross-entropy with probability targets (torch semantics:
+// loss = mean_i [ -sum_c t_ic * log_softmax(y_i)_c ]).
+// One wave per row; lanes stride over C; wave shuffle reductions (wave=64).
+// Saves probs and per-row target sums for the backward.
+// ---------------------------------------------------------------------------
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+__global__ void k_ce_fwd_f32(const float* __restrict__ Y,
+                             const float* __restrict__ T,
+                             float* __restrict__ P,
+                             float* __restrict__ tsum,
+                             float* __restrict__ loss,  // pre-zeroed scalar
+                             int B, int C) {
+  const int lane = threadIdx.x & 63;
+  const int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
+  if (row >= B) return;
+  const float* y = Y + (size_t)row * C;
+  const float* t = T + (size_t)row * C;
+  float* p = P + (size_t)row * C;
+  float m = -INFINITY;
+  for (int c = lane; c < C; c += 64) m = fmaxf(m, y[c]);
+  m = wave_max(m);
+  float se = 0.f, ts = 0.f, ty = 0.f;
+  for (int c = lane; c < C; c += 64) {
+    const float e = __expf(y[c] - m);
+    se += e; ts += t[c]; ty += t[c] * y[c];
+  }
+  se = wave_sum(se); ts = wave_sum(ts); ty = wave_sum(ty);
+  const float inv_se = 1.f / se;
+  for (int c = lane; c < C; c += 64) p[c] = __expf(y[c] - m) * inv_se;
+  if (lane == 0) {
+    tsum[row] = ts;
+    const float logZ = m + __logf(se);
+    atomicAdd(loss, (ts * logZ - ty) / (float)B);
+  }
+}
+
+std::vector<torch::Tensor> ce_fwd(torch::Tensor y, torch::Tensor t) {
+  auto yc = y.contiguous();
+  auto tc = t.contiguous();
+  const int B = (int)yc.size(0), C = (int)yc.size(1);
+  auto probs = at::empty_like(yc);
+  auto tsum = at::empty({B}, yc.options());
+  auto loss = at::zeros({}, yc.options());
+  dim3 grid(cdiv(B, 4));
+  hipLaunchKernelGGL(k_ce_fwd_f32, grid, dim3(256), 0, cur_stream(),
+                     yc.data_ptr<float>(), tc.data_ptr<float>(),
+                     probs.data_ptr<float>(), tsum.data_ptr<float>(),
+                     loss.data_ptr<float>(), B, C);
+  HIP_OK(hipGetLastError());
+  return {loss, probs, tsum};
+}
+
+// dY = (tsum_row * p - t) * grad_scale / B
+__global__ void k_ce_bwd_f32(const float* __restrict__ P,
+                             const float* __restrict__ T,
+                             const float* __restrict__ tsum,
+                             float* __restrict__ dY,
+                             float scale, int B, int C) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t n = (int64_t)B * C;
+  if (i >= n) return;
+  const int row = (int)(i / C);
+  dY[i] = (tsum[row] * P[i] - T[i]) * scale;
+}
+
+torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor t,
+                     torch::Tensor tsum, double grad_scale) {
+  const int B = (int)probs.size(0), C = (int)probs.size(1);
+  auto dy = at::empty_like(probs);
+  const int64_t n = (int64_t)B * C;
+  hipLaunchKernelGGL(k_ce_bwd_f32, dim3(cdiv(n, 256)), dim3(256), 0, cur_stream(),
+                     probs.data_ptr<float>(), t.contiguous().data_ptr<float>(),
+                     tsum.data_ptr<float>(), dy.data_ptr<float>(),
+                     (float)(grad_scale / B), B, C);
+  HIP_OK(hipGetLastError());
+  return dy;
+}
+
+// ---------------------------------------------------------------------------
+// MSE: loss = mean((y - t)^2); dY = 2 (y - t) * grad_scale / numel
+// ---------------------------------------------------------------------------
+__global__ void k_mse_fwd_f32(const float* __restrict__ Y,
+                              const float* __restrict__ T,
+                              float* __restrict__ loss,  // pre-zeroed
+                              int64_t n, float inv_n) {
+  float acc = 0.f;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float d = Y[i] - T[i];
+    acc += d * d;
+  }
+  acc = wave_sum(acc);
+  __shared__ float warp_part[4];
+  const int wave = threadIdx.x >> 6;
+  if ((threadIdx.x & 63) == 0) warp_part[wave] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.f;
+    for (int w = 0; w < (int)(blockDim.x >> 6); ++w) s += warp_part[w];
+    atomicAdd(loss, s * inv_n);
+  }
+}
+
+torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t) {
+  auto yc = y.contiguous();
+  auto tc = t.contiguous();
+  const int64_t n = yc.numel();
+  auto loss = at::zeros({}, yc.options());
+  const int blocks = (int)std::min<int64_t>(cdiv(n, 256), 2048);
+  hipLaunchKernelGGL(k_mse_fwd_f32, dim3(blocks), dim3(256), 0, cur_stream(),
+                     yc.data_ptr<float>(), tc.data_ptr<float>(),
+                     loss.data_ptr<float>(), n, 1.f / (float)n);
+  HIP_OK(hipGetLastError());
+  return loss;
+}
+
+__global__ void k_mse_bwd_f32(const float* __restrict__ Y,
+                              const float* __restrict__ T,
+                              float* __restrict__ dY, float scale, int64_t n) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dY[i] = (Y[i] - T[i]) * scale;
+}
+
+torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale) {
+  auto yc = y.contiguous();
+  auto tc = t.contiguous();
+  const int64_t n = yc.numel();
+  auto dy = at::empty_like(yc);
+  hipLaunchKernelGGL(k_mse_bwd_f32, dim3(cdiv(n, 256)), dim3(256), 0, cur_stream(),
+                     yc.data_ptr<float>(), tc.data_ptr<float>(),
+                     dy.data_ptr<float>(), (float)(2.0 * grad_scale / n), n);
+  HIP_OK(hipGetLastError());
+  return dy;
+}
+
+// ---------------------------------------------------------------------------
+// Fused SGD over a flat bucket (+ fold zero_grad): p -= lr * g; g = 0.
+// Vectorized float4 (16 B/lane); bucket lengths are padded to a multiple of
+// 4 elements by the reducer, so the f32x4 path covers everything.
+// ---------------------------------------------------------------------------
+__global__ void k_sgd_flat_f32(float4* __restrict__ p, float4* __restrict__ g,
+                               float lr, int64_t n4, int zero) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  float4 gv = g[i];
+  float4 pv = p[i];
+  pv.x -= lr * gv.x; pv.y -= lr * gv.y; pv.z -= lr * gv.z; pv.w -= lr * gv.w;
+  p[i] = pv;
+  if (zero) g[i] = make_float4(0.f, 0.f, 0.f, 0.f);
+}
+
+void sgd_flat(torch::Tensor param_flat, torch::Tensor grad_flat,
+              double lr, bool zero_grad) {
+  TORCH_CHECK(param_flat.is_cuda() && param_flat.is_contiguous() &&
+              grad_flat.is_contiguous(), "flat buffers must be contiguous");
+  TORCH_CHECK(param_flat.numel() % 4 == 0,
+              "bucket length must be padded to a multiple of 4");
+  const int64_t n4 = param_flat.numel() / 4;
+  hipLaunchKernelGGL(k_sgd_flat_f32, dim3(cdiv(n4, 256)), dim3(256), 0,
+                     cur_stream(),
+                     reinterpret_cast<float4*>(param_flat.data_ptr<float>()),
+                     reinterpret_cast<float4*>(grad_flat.data_ptr<float>()),
+                     (float)lr, n4, zero_grad ? 1 : 0);
+  HIP_OK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// Bucket flatten/unflatten: one launch per bucket moves all
+// member tensors via a precomputed per-block copy plan.
+// plan row (int64 x3): [src_addr, bucket_elem_offset, numel] for each block's
+// chunk of ELEMS_PER_BLOCK elements. Tensor bucket offsets are 4-elem
+// aligned (reducer), torch allocations are 256-B aligned, so float4 moves
+// cover whole chunks; the (rare) tail of each tensor goes scalar.
+// ---------------------------------------------------------------------------
+static constexpr int64_t ELEMS_PER_BLOCK = 2048;  // 256 threads x 8 f32
+
+template <bool INTO_BUCKET>
+__global__ void k_bucket_copy(float* __restrict__ bucket,
+                              const int64_t* __restrict__ plan,
+                              int zero_src) {
+  const int64_t* row = plan + (int64_t)blockIdx.x * 3;
+  float* tp = reinterpret_cast<float*>(row[0]);  // tensor chunk
+  float* bp = bucket + row[1];                   // bucket segment
+  const int64_t n = row[2];
+  float* src = INTO_BUCKET ? tp : bp;
+  float* dst = INTO_BUCKET ? bp : tp;
+  const int64_t n4 = n >> 2;
+  float4* s4 = reinterpret_cast<float4*>(src);
+  float4* d4 = reinterpret_cast<float4*>(dst);
+  const float4 z4 = make_float4(0.f, 0.f, 0.f, 0.f);
+  for (int64_t i = threadIdx.x; i < n4; i += blockDim.x) {
+    d4[i] = s4[i];
+    if (zero_src) s4[i] = z4;  // fold zero_grad into the gather
+  }
+  for (int64_t i = (n4 << 2) + threadIdx.x; i < n; i += blockDim.x) {
+    dst[i] = src[i];
+    if (zero_src) src[i] = 0.f;
+  }
+}
+
+torch::Tensor build_copy_plan(const std::vector<torch::Tensor>& tensors,
+                              const std::vector<int64_t>& offsets,
+                              torch::Device device) {
+  std::vector<int64_t> rows;
+  for (size_t t = 0; t < tensors.size(); ++t) {
+    TORCH_CHECK(tensors[t].is_contiguous(), "bucket members must be contiguous");
+    const int64_t numel = tensors[t].numel();
+    const int64_t addr = (int64_t)(uintptr_t)tensors[t].data_ptr();
+    for (int64_t base = 0; base < numel; base += ELEMS_PER_BLOCK) {
+      rows.push_back(addr + base * (int64_t)sizeof(float));
+      rows.push_back(offsets[t] + base);
+      rows.push_back(std::min(ELEMS_PER_BLOCK, numel - base));
+    }
+  }
+  auto plan = torch::from_blob(rows.data(), {(int64_t)rows.size() / 3, 3},
+                               torch::kInt64).clone();
+  return plan.to(device);
+}
+
+void flatten_into(torch::Tensor bucket, torch::Tensor plan,
+                  int64_t total_blocks, bool zero_src) {
+  hipLaunchKernelGGL((k_bucket_copy<true>), dim3((uint32_t)total_blocks),
+                     dim3(256), 0, cur_stream(),
+                     bucket.data_ptr<float>(), plan.data_ptr<int64_t>(),
+                     zero_src ? 1 : 0);
+  HIP_OK(hipGetLastError());
+}
+
+void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_blocks) {
+  hipLaunchKernelGGL((k_bucket_copy<false>), dim3((uint32_t)total_blocks),
+                     dim3(256), 0, cur_stream(),
+                     bucket.data_ptr<float>(), plan.data_ptr<int64_t>(), 0);
+  HIP_OK(hipGetLastError());
+}
+
+// ---------------------------------------------------------------------------
+// Fused toy training step (fwd + loss-grad + bwd in ONE kernel): the
+// reference hot loop single_gpu.py:21-26 for model = Linear(K,1), minus the
+// optimizer (which needs the all-reduced grads). Single workgroup, 64
+// threads (one wave): at 84 B of gradients the step is launch-latency
+// bound, so one launch replaces five (SURVEY §7 hard-part 2).
+// Layout: param_flat = [w(K) | b | pad], grad_flat same.
+// Supports B <= 128, K <= 32.
+// ---------------------------------------------------------------------------
+__global__ void k_toy_fused_f32(const float* __restrict__ X,
+                                const float* __restrict__ T,
+                                const float* __restrict__ param,
+                                float* __restrict__ grad,
+                                float* __restrict__ loss_out,
+                                int B, int K, int use_mse) {
+  const int lane = threadIdx.x;
+  const int r = lane & 15, q = lane >> 4;
+  __shared__ float dy_s[128];
+
+  // forward: y_i = sum_k X[i,k] w_k + b ; MFMA tiles of 16 rows, j=0 column.
+  const float bterm = param[K];
+  const int ntile = (B + 15) / 16;
+  float dy_own = 0.f;  // this lane's dY rows (written via LDS below)
+  float loss_acc = 0.f;
+  for (int tm = 0; tm < ntile; ++tm) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int m = tm * 16 + r;
+    for (int k0 = 0; k0 < K; k0 += 4) {
+      const int k = k0 + q;
+      const float a = (m < B && k < K) ? X[(size_t)m * K + k] : 0.f;
+      const float b = (r == 0 && k < K) ? param[k] : 0.f;  // B[k][j=0]
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+    if (r == 0) {  // lanes 0,16,32,48 hold col j=0; rows q*4+i
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = tm * 16 + q * 4 + i;
+        if (row < B) {
+          const float y = acc[i] + bterm;
+          const float t = T[row];
+          float dy;
+          if (use_mse) {
+            const float d = y - t;
+            loss_acc += d * d;
+            dy = 2.f * d / (float)B;
+          } else {
+            // CE over one logit: log_softmax == 0 -> loss == 0, dY == 0
+            // (the reference's degenerate loss).
+            dy = 0.f;
+          }
+          dy_s[row] = dy;
+        }
+      }
+    }
+  }
+  __syncthreads();
+
+  // backward: dw_k = sum_i dY_i X[i,k] via MFMA (A row j=0 = dY), db = sum dY
+  const int ktiles = (K + 15) / 16;
+  for (int tk = 0; tk < ktiles; ++tk) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int k = tk * 16 + r;
+    for (int i0 = 0; i0 < B; i0 += 4) {
+      const int i = i0 + q;
+      const float a = (r == 0 && i < B) ? dy_s[i] : 0.f;  // A[j=0][i]
+      const float b = (i < B && k < K) ? X[(size_t)i * K + k] : 0.f;
+      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+    }
+    // D row j=0 lives in reg 0 of lanes with q==0; col = k
+    if (q == 0 && k < K) grad[k] = acc[0];
+  }
+  // db + loss reduce
+  float dbp = 0.f;
+  for (int i = lane; i < B; i += 64) dbp += dy_s[i];
+  dbp = wave_sum(dbp);
+  if (use_mse) loss_acc = wave_sum(loss_acc);
+  if (lane == 0) {
+    grad[K] = dbp;
+    if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
+  }
+}
+
+void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
+                       torch::Tensor param_flat, torch::Tensor grad_flat,
+                       torch::Tensor loss_out, bool use_mse) {
+  const int B = (int)x.size(0), K = (int)x.size(1);
+  TORCH_CHECK(B <= 128 && K <= 32, "toy fused kernel supports B<=128, K<=32");
+  TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
+  hipLaunchKernelGGL(k_toy_fused_f32, dim3(1), dim3(64), 0, cur_stream(),
+                     x.data_ptr<float>(), t.data_ptr<float>(),
+                     param_flat.data_ptr<float>(), grad_flat.data_ptr<float>(),
+                     loss_out.defined() ? loss_out.data_ptr<float>() : nullptr,
+                     B, K, use_mse ? 1 : 0);
+  HIP_OK(hipGetLastError());
+}
+
+}  // namespace mi355x

@@ -1,0 +1,104 @@
+// See rccl_comm.h. Native HIP + RCCL; no CUDA compatibility paths.
+#include "rccl_comm.h"
+
+#include <ATen/hip/HIPContext.h>
+#include <c10/hip/HIPStream.h>
+
+#define HIP_OK(expr)                                                          \
+  do {                                                                        \
+    hipError_t _e = (expr);                                                   \
+    TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));      \
+  } while (0)
+
+#define NCCL_OK(expr)                                                         \
+  do {                                                                        \
+    ncclResult_t _r = (expr);                                                 \
+    TORCH_CHECK(_r == ncclSuccess, "RCCL error: ", ncclGetErrorString(_r));   \
+  } while (0)
+
+namespace mi355x {
+
+static ncclDataType_t nccl_dtype(const torch::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kHalf: return ncclFloat16;
+    case at::kDouble: return ncclFloat64;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    default: TORCH_CHECK(false, "unsupported dtype for RCCL: ", t.scalar_type());
+  }
+}
+
+static hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+std::string RcclComm::make_unique_id() {
+  ncclUniqueId id;
+  NCCL_OK(ncclGetUniqueId(&id));
+  return std::string(id.internal, NCCL_UNIQUE_ID_BYTES);
+}
+
+RcclComm::RcclComm(const std::string& unique_id, int rank, int world, int device)
+    : rank_(rank), world_(world), device_(device) {
+  TORCH_CHECK((int)unique_id.size() == NCCL_UNIQUE_ID_BYTES,
+              "bad unique id size ", unique_id.size());
+  ncclUniqueId id;
+  memcpy(id.internal, unique_id.data(), NCCL_UNIQUE_ID_BYTES);
+  HIP_OK(hipSetDevice(device));
+  // Dedicated, high-priority comm stream: bucket all-reduces launched here
+  // overlap with backward compute on the torch stream.
+  int least = 0, greatest = 0;
+  HIP_OK(hipDeviceGetStreamPriorityRange(&least, &greatest));
+  HIP_OK(hipStreamCreateWithPriority(&comm_stream_, hipStreamNonBlocking, greatest));
+  HIP_OK(hipEventCreateWithFlags(&ready_ev_, hipEventDisableTiming));
+  HIP_OK(hipEventCreateWithFlags(&done_ev_, hipEventDisableTiming));
+  NCCL_OK(ncclCommInitRank(&comm_, world, id, rank));
+}
+
+RcclComm::~RcclComm() {
+  if (comm_) ncclCommDestroy(comm_);
+  if (comm_stream_) (void)hipStreamDestroy(comm_stream_);
+  if (ready_ev_) (void)hipEventDestroy(ready_ev_);
+  if (done_ev_) (void)hipEventDestroy(done_ev_);
+}
+
+void RcclComm::all_reduce_avg(torch::Tensor t) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous(), "all_reduce needs contiguous device tensor");
+  // order: comm stream waits for the producer (compute) stream ...
+  HIP_OK(hipEventRecord(ready_ev_, cur_stream()));
+  HIP_OK(hipStreamWaitEvent(comm_stream_, ready_ev_, 0));
+  NCCL_OK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(), nccl_dtype(t),
+                        ncclAvg, comm_, comm_stream_));
+  // ... and remembers completion so join_compute() can fence the optimizer.
+  HIP_OK(hipEventRecord(done_ev_, comm_stream_));
+  any_inflight_ = true;
+}
+
+void RcclComm::all_reduce_avg_inline(torch::Tensor t) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous());
+  NCCL_OK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(), nccl_dtype(t),
+                        ncclAvg, comm_, cur_stream()));
+}
+
+void RcclComm::broadcast(torch::Tensor t, int root) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous());
+  NCCL_OK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(), nccl_dtype(t),
+                        root, comm_, cur_stream()));
+}
+
+void RcclComm::join_compute() {
+  if (!any_inflight_) return;
+  HIP_OK(hipStreamWaitEvent(cur_stream(), done_ev_, 0));
+  any_inflight_ = false;
+}
+
+void RcclComm::barrier() {
+  auto t = at::zeros({1}, at::TensorOptions().device(at::kCUDA, device_)
+                              .dtype(at::kFloat));
+  all_reduce_avg_inline(t);
+  HIP_OK(hipStreamSynchronize(cur_stream()));
+}
+
+}  // namespace mi355x

@@ -1,0 +1,119 @@
+"""Communicator layer: RCCL-over-xGMI driven directly, with a gloo shadow
+for CPU-only testing.
+
+The reference delegates all communication to c10d ProcessGroupNCCL
+(init sites multigpu.py:20, multigpu_torchrun.py:13,
+multinode_torchrun.py:13, multigpu_profile.py:97). Here torch.distributed
+is used ONLY for rendezvous (rank discovery + the TCP store that carries
+the 128-byte RCCL unique id); every gradient byte moves
+through the framework's own `RcclComm` (ops/csrc/rccl_comm.hip) on a
+dedicated comm stream.
+
+`GlooComm` mirrors the same interface over torch.distributed's gloo backend
+so the reducer's bucketing/overlap logic is exercised by multi-process CPU
+tests (SURVEY §4 consequence (a)).
+"""
+
+from __future__ import annotations
+
+import base64
+import os
+from typing import List, Optional
+
+import torch
+import torch.distributed as dist
+
+
+def ddp_setup(rank: Optional[int] = None, world_size: Optional[int] = None,
+              backend: Optional[str] = None) -> None:
+    """Process-group bootstrap covering both reference styles:
+
+    - explicit rank/world (mp.spawn style, reference multigpu.py:12-20:
+      sets MASTER_ADDR/PORT defaults then init_process_group)
+    - env-var style under torchrun (reference multigpu_torchrun.py:12-13).
+
+    The backend defaults to "nccl" on GPU hosts (the string is kept for the
+    torchrun contract — tensor traffic still goes through RcclComm) and
+    "gloo" on CPU-only hosts.
+    """
+    if backend is None:
+        backend = "nccl" if torch.cuda.is_available() else "gloo"
+    if rank is not None and world_size is not None:
+        os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+        os.environ.setdefault("MASTER_PORT", "12355")
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
+    else:
+        dist.init_process_group(backend)
+    if torch.cuda.is_available():
+        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank or 0)))
+
+
+class GlooComm:
+    """CPU shadow communicator: same interface as the RCCL path, backed by
+    torch.distributed gloo. Async handles model the comm-stream overlap."""
+
+    def __init__(self):
+        assert dist.is_initialized()
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        self._handles: List = []
+
+    def all_reduce_avg(self, t: torch.Tensor) -> None:
+        h = dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=True)
+        self._handles.append((h, t))
+
+    def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
+        dist.broadcast(t, src=root)
+
+    def join_compute(self) -> None:
+        for h, t in self._handles:
+            h.wait()
+            t.div_(self.world)
+        self._handles.clear()
+
+    def barrier(self) -> None:
+        dist.barrier()
+
+
+class RcclCommAdapter:
+    """Thin adapter over the native RcclComm (exchange of the unique id via
+    the c10d rendezvous store, then direct RCCL)."""
+
+    _KEY = "mi355x_ddp/rccl_uid"
+
+    def __init__(self, device: torch.device):
+        from .. import ops
+        assert dist.is_initialized()
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        store = dist.distributed_c10d._get_default_store()
+        if self.rank == 0:
+            uid = ops.ext().RcclComm.make_unique_id()
+            store.set(self._KEY, base64.b64encode(uid).decode())
+        uid = base64.b64decode(store.get(self._KEY))
+        self._comm = ops.ext().RcclComm(uid, self.rank, self.world,
+                                        device.index or 0)
+
+    def all_reduce_avg(self, t: torch.Tensor) -> None:
+        self._comm.all_reduce_avg(t)
+
+    def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
+        self._comm.broadcast(t, root)
+
+    def join_compute(self) -> None:
+        self._comm.join_compute()
+
+    def barrier(self) -> None:
+        self._comm.barrier()
+
+
+def create_comm(device: torch.device):
+    """Pick the communicator for this process, or None when world size is 1
+    (single-process runs use the same reducer with no collectives)."""
+    if not (dist.is_available() and dist.is_initialized()):
+        return None
+    if dist.get_world_size() == 1:
+        return None
+    if device.type == "cuda":
+        return RcclCommAdapter(device)
+    return GlooComm()

@@ -1,0 +1,64 @@
+"""DistributedDataParallel replacement built on the native Reducer + RCCL.
+
+API parity with the reference's DDP usage:
+- `DDP(model)` at wrap time broadcasts parameters and buffers from rank 0
+  (reference wrap sites multigpu.py:36, multigpu_torchrun.py:34,
+  multinode_torchrun.py:35, multigpu_profile.py:41; broadcast = N4)
+- `.module` exposes the wrapped model for checkpoint unwrapping
+  (reference multigpu.py:54 `model.module.state_dict()`)
+- the per-step contract is loss.backward() -> finalize_backward() ->
+  optimizer.step() (the Trainer calls finalize_backward; stock DDP hides
+  the equivalent fence inside its autograd hooks)
+
+Also usable with world_size == 1 (no communicator): the model still gets
+flat parameter/grad buckets and the fused SGD path, which is how the
+single-GPU benchmark runs the same engine.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+
+from .comm import create_comm
+from .reducer import Reducer
+
+
+class DDP(torch.nn.Module):
+    def __init__(self, module: torch.nn.Module,
+                 bucket_cap_mb: Optional[float] = None,
+                 comm=None, grad_views: Optional[bool] = None):
+        super().__init__()
+        self.module = module
+        params = list(module.parameters())
+        device = params[0].device if params else torch.device("cpu")
+        self.comm = comm if comm is not None else create_comm(device)
+        self.reducer = Reducer(params, comm=self.comm,
+                               bucket_cap_mb=bucket_cap_mb,
+                               grad_views=grad_views)
+        # wrap-time module-state sync: params coalesced per bucket,
+        # buffers (e.g. BN running stats) individually
+        if self.comm is not None:
+            self.reducer.broadcast_params(root=0)
+            for buf in module.buffers():
+                if not buf.numel():
+                    continue
+                if buf.is_contiguous():
+                    self.comm.broadcast(buf, 0)
+                else:
+                    tmp = buf.contiguous()
+                    self.comm.broadcast(tmp, 0)
+                    with torch.no_grad():
+                        buf.copy_(tmp)
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    def finalize_backward(self) -> None:
+        """Launch straggler buckets and fence compute on the comm stream.
+        Must be called between loss.backward() and optimizer.step()."""
+        self.reducer.finalize()
+
+    # parity helper: stock DDP state_dict carries the "module." prefix;
+    # nn.Module gives us that for free since `module` is a submodule.

@@ -1,0 +1,27 @@
+"""Env-var knobs (the reference's only config channels are positional argv
+and env vars — SURVEY §5.6; internal engine knobs live here).
+
+MI355X_BUCKET_MB        gradient bucket cap (default 25)
+MI355X_FIRST_BUCKET_MB  first-bucket cap so the first all-reduce fires
+                        early (default 1)
+MI355X_GRAD_VIEWS       1 = grads are views into flat buckets (zero-copy,
+                        default); 0 = flatten-kernel gather per bucket
+MI355X_HIPGRAPH         1 = capture the steady-state train step in a
+                        hipGraph and replay it (bench fast path)
+"""
+
+import os
+
+
+def env_flag(name: str, default: bool = False) -> bool:
+    v = os.environ.get(name)
+    if v is None:
+        return default
+    return v not in ("0", "false", "False", "")
+
+
+def env_float(name: str, default: float) -> float:
+    try:
+        return float(os.environ.get(name, default))
+    except ValueError:
+        return default
