@@ -468,13 +468,16 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
                                 const float* __restrict__ param,
                                 float* __restrict__ grad,
                                 float* __restrict__ loss_out,
-                                int B, int K, int use_mse) {
+                                int B, int K, int use_mse,
+                                int w_off, int b_off) {
   const int lane = threadIdx.x;
   const int r = lane & 15, q = lane >> 4;
   __shared__ float dy_s[128];
 
   // forward: y_i = sum_k X[i,k] w_k + b ; MFMA tiles of 16 rows, j=0 column.
-  const float bterm = param[K];
+  // w/b live at arbitrary offsets inside the reducer's flat bucket
+  // (reverse-registration order puts bias first).
+  const float bterm = param[b_off];
   const int ntile = (B + 15) / 16;
   float dy_own = 0.f;  // this lane's dY rows (written via LDS below)
   float loss_acc = 0.f;
@@ -484,7 +487,7 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
     for (int k0 = 0; k0 < K; k0 += 4) {
       const int k = k0 + q;
       const float a = (m < B && k < K) ? X[(size_t)m * K + k] : 0.f;
-      const float b = (r == 0 && k < K) ? param[k] : 0.f;  // B[k][j=0]
+      const float b = (r == 0 && k < K) ? param[w_off + k] : 0.f;  // B[k][j=0]
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
     if (r == 0) {  // lanes 0,16,32,48 hold col j=0; rows q*4+i
@@ -523,7 +526,7 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
     // D row j=0 lives in reg 0 of lanes with q==0; col = k
-    if (q == 0 && k < K) grad[k] = acc[0];
+    if (q == 0 && k < K) grad[w_off + k] = acc[0];
   }
   // db + loss reduce
   float dbp = 0.f;
@@ -531,14 +534,15 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
   dbp = wave_sum(dbp);
   if (use_mse) loss_acc = wave_sum(loss_acc);
   if (lane == 0) {
-    grad[K] = dbp;
+    grad[b_off] = dbp;
     if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
   }
 }
 
 void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor param_flat, torch::Tensor grad_flat,
-                       torch::Tensor loss_out, bool use_mse) {
+                       torch::Tensor loss_out, bool use_mse,
+                       int64_t w_off, int64_t b_off) {
   const int B = (int)x.size(0), K = (int)x.size(1);
   TORCH_CHECK(B <= 128 && K <= 32, "toy fused kernel supports B<=128, K<=32");
   TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
@@ -546,7 +550,7 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                      x.data_ptr<float>(), t.data_ptr<float>(),
                      param_flat.data_ptr<float>(), grad_flat.data_ptr<float>(),
                      loss_out.defined() ? loss_out.data_ptr<float>() : nullptr,
-                     B, K, use_mse ? 1 : 0);
+                     B, K, use_mse ? 1 : 0, (int)w_off, (int)b_off);
   HIP_OK(hipGetLastError());
 }
 

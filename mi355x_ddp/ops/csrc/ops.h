@@ -52,6 +52,7 @@ void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_bloc
 // toy path, SURVEY §7 hard-part 2).
 void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor param_flat, torch::Tensor grad_flat,
-                       torch::Tensor loss_out, bool use_mse);
+                       torch::Tensor loss_out, bool use_mse,
+                       int64_t w_off, int64_t b_off);
 
 }  // namespace mi355x

@@ -62,6 +62,10 @@ class GlooComm:
         h = dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=True)
         self._handles.append((h, t))
 
+    def all_reduce_avg_inline(self, t: torch.Tensor) -> None:
+        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        t.div_(self.world)
+
     def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
         dist.broadcast(t, src=root)
 
@@ -96,6 +100,9 @@ class RcclCommAdapter:
 
     def all_reduce_avg(self, t: torch.Tensor) -> None:
         self._comm.all_reduce_avg(t)
+
+    def all_reduce_avg_inline(self, t: torch.Tensor) -> None:
+        self._comm.all_reduce_avg_inline(t)
 
     def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
         self._comm.broadcast(t, root)

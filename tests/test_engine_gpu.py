@@ -1,0 +1,85 @@
+"""GPU tests: the fused/graphed toy engines train identically to the
+generic autograd+reducer path (same kernels, one launch instead of five)."""
+
+import json
+import os
+import subprocess
+import sys
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+LR = 0.05
+STEPS = 20
+
+
+def _data(seed=9):
+    g = torch.Generator().manual_seed(seed)
+    return (torch.rand(STEPS, 32, 20, generator=g).to(DEV),
+            torch.rand(STEPS, 32, 1, generator=g).to(DEV))
+
+
+def _train_autograd():
+    from mi355x_ddp import ops
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import DDP, FusedSGD
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(DEV)
+    engine = DDP(model)
+    opt = FusedSGD(model.parameters(), lr=LR)
+    opt.attach_reducer(engine.reducer)
+    X, T = _data()
+    for s in range(STEPS):
+        loss = ops.mse_loss(engine(X[s]), T[s])
+        loss.backward()
+        engine.finalize_backward()
+        opt.step()
+    return model.weight.detach().cpu().clone(), model.bias.detach().cpu().clone()
+
+
+def _train_fused(graphed: bool):
+    from mi355x_ddp.engine import GraphedToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(DEV)
+    cls = GraphedToyStep if graphed else ToyFusedStep
+    eng = cls(model, comm=None, lr=LR, use_mse=True)
+    X, T = _data()
+    for s in range(STEPS):
+        eng.step(X[s], T[s])
+    torch.cuda.synchronize()
+    return model.weight.detach().cpu().clone(), model.bias.detach().cpu().clone()
+
+
+def test_fused_matches_autograd():
+    w_ref, b_ref = _train_autograd()
+    w, b = _train_fused(graphed=False)
+    assert torch.allclose(w, w_ref, atol=1e-5), (w - w_ref).abs().max()
+    assert torch.allclose(b, b_ref, atol=1e-5)
+
+
+def test_graphed_matches_fused():
+    w_ref, b_ref = _train_fused(graphed=False)
+    w, b = _train_fused(graphed=True)
+    # the graphed path runs the same kernels; capture warmup runs one extra
+    # real step during capture, so re-run reference with the same schedule:
+    # GraphedToyStep's first step() performs capture (1 warmup step + no
+    # replay), i.e. step 0 is applied once either way -> identical history.
+    assert torch.allclose(w, w_ref, atol=1e-5), (w - w_ref).abs().max()
+    assert torch.allclose(b, b_ref, atol=1e-5)
+
+
+def test_bench_single_gpu_json():
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(root, "bench.py"), "--steps", "50",
+         "--warmup", "10", "--p50-probes", "8"],
+        cwd=root, capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    line = out.stdout.strip().splitlines()[-1]
+    r = json.loads(line)
+    assert r["n_gpus"] == 1 and r["steps"] == 50
+    assert r["value"] > 0 and r["dtype"] == "fp32"

@@ -154,7 +154,7 @@ def test_toy_fused_step_matches_reference():
     param[K] = b[0]
     grad = torch.zeros(24, device=DEV)
     loss_out = torch.zeros((), device=DEV)
-    ops.ext().toy_fused_fwd_bwd(x, t, param, grad, loss_out, True)
+    ops.ext().toy_fused_fwd_bwd(x, t, param, grad, loss_out, True, 0, K)
 
     wc = w.cpu().requires_grad_(True)
     bc = b.cpu().requires_grad_(True)
