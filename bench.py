@@ -33,7 +33,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=200)
     p.add_argument("--engine", choices=["fused", "graph", "autograd"],
-                   default="graph")
+                   default="fused")
     p.add_argument("--batch", type=int, default=32, help="batch per rank")
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--dataset", type=int, default=2048)

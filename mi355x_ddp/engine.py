@@ -55,11 +55,17 @@ class ToyFusedStep:
                          if track_loss else torch.Tensor())
 
     def step(self, x: torch.Tensor, t: torch.Tensor) -> None:
+        if self.comm is None:
+            # single-process: one launch = one full training step
+            # (fwd + loss grad + bwd + SGD fused; no collective to wait for)
+            ops.ext().toy_fused_fwd_bwd(x, t, self.flat_param, self.flat_grad,
+                                        self.loss_out, self.use_mse,
+                                        self.w_off, self.b_off, self.lr)
+            return
         ops.ext().toy_fused_fwd_bwd(x, t, self.flat_param, self.flat_grad,
                                     self.loss_out, self.use_mse,
-                                    self.w_off, self.b_off)
-        if self.comm is not None:
-            self.comm.all_reduce_avg_inline(self.flat_grad)
+                                    self.w_off, self.b_off, 0.0)
+        self.comm.all_reduce_avg_inline(self.flat_grad)
         ops.ext().sgd_flat(self.flat_param, self.flat_grad, self.lr, True)
 
 

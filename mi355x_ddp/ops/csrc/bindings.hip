@@ -33,7 +33,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("toy_fused_fwd_bwd", &mi355x::toy_fused_fwd_bwd, py::arg("x"),
         py::arg("t"), py::arg("param_flat"), py::arg("grad_flat"),
         py::arg("loss_out"), py::arg("use_mse") = true,
-        py::arg("w_off") = 0, py::arg("b_off") = 0);
+        py::arg("w_off") = 0, py::arg("b_off") = 0, py::arg("lr") = 0.0);
 
   py::class_<mi355x::RcclComm>(m, "RcclComm")
       .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),

@@ -465,11 +465,11 @@ void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_bloc
 // ---------------------------------------------------------------------------
 __global__ void k_toy_fused_f32(const float* __restrict__ X,
                                 const float* __restrict__ T,
-                                const float* __restrict__ param,
+                                float* __restrict__ param,
                                 float* __restrict__ grad,
                                 float* __restrict__ loss_out,
                                 int B, int K, int use_mse,
-                                int w_off, int b_off) {
+                                int w_off, int b_off, float lr) {
   const int lane = threadIdx.x;
   const int r = lane & 15, q = lane >> 4;
   __shared__ float dy_s[128];
@@ -525,8 +525,13 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
       const float b = (i < B && k < K) ? X[(size_t)i * K + k] : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
-    // D row j=0 lives in reg 0 of lanes with q==0; col = k
-    if (q == 0 && k < K) grad[w_off + k] = acc[0];
+    // D row j=0 lives in reg 0 of lanes with q==0; col = k.
+    // lr > 0: world-size-1 fast path — apply SGD in-kernel (no all-reduce
+    // needed, grads never materialize; one launch = one training step).
+    if (q == 0 && k < K) {
+      if (lr > 0.f) param[w_off + k] -= lr * acc[0];
+      else grad[w_off + k] = acc[0];
+    }
   }
   // db + loss reduce
   float dbp = 0.f;
@@ -534,7 +539,8 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
   dbp = wave_sum(dbp);
   if (use_mse) loss_acc = wave_sum(loss_acc);
   if (lane == 0) {
-    grad[b_off] = dbp;
+    if (lr > 0.f) param[b_off] -= lr * dbp;
+    else grad[b_off] = dbp;
     if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
   }
 }
@@ -542,7 +548,7 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
 void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor param_flat, torch::Tensor grad_flat,
                        torch::Tensor loss_out, bool use_mse,
-                       int64_t w_off, int64_t b_off) {
+                       int64_t w_off, int64_t b_off, double lr) {
   const int B = (int)x.size(0), K = (int)x.size(1);
   TORCH_CHECK(B <= 128 && K <= 32, "toy fused kernel supports B<=128, K<=32");
   TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
@@ -550,7 +556,7 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                      x.data_ptr<float>(), t.data_ptr<float>(),
                      param_flat.data_ptr<float>(), grad_flat.data_ptr<float>(),
                      loss_out.defined() ? loss_out.data_ptr<float>() : nullptr,
-                     B, K, use_mse ? 1 : 0, (int)w_off, (int)b_off);
+                     B, K, use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr);
   HIP_OK(hipGetLastError());
 }
 

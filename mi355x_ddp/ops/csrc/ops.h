@@ -50,9 +50,11 @@ void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_bloc
 // grads (dw|db) for Linear(K,1), computes fwd+MSE-or-CE loss grad+bwd and
 // writes gradients into the bucket in ONE kernel launch (the latency-bound
 // toy path, SURVEY §7 hard-part 2).
+// lr > 0: apply the SGD update in-kernel (world-size-1 single-launch step);
+// lr <= 0: write gradients into grad_flat for the all-reduce path.
 void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor param_flat, torch::Tensor grad_flat,
                        torch::Tensor loss_out, bool use_mse,
-                       int64_t w_off, int64_t b_off);
+                       int64_t w_off, int64_t b_off, double lr);
 
 }  // namespace mi355x

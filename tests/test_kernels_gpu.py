@@ -154,7 +154,7 @@ def test_toy_fused_step_matches_reference():
     param[K] = b[0]
     grad = torch.zeros(24, device=DEV)
     loss_out = torch.zeros((), device=DEV)
-    ops.ext().toy_fused_fwd_bwd(x, t, param, grad, loss_out, True, 0, K)
+    ops.ext().toy_fused_fwd_bwd(x, t, param, grad, loss_out, True, 0, K, 0.0)
 
     wc = w.cpu().requires_grad_(True)
     bc = b.cpu().requires_grad_(True)
@@ -164,6 +164,50 @@ def test_toy_fused_step_matches_reference():
     assert torch.allclose(loss_out.cpu(), loss, atol=1e-5)
     assert torch.allclose(grad[:K].cpu(), wc.grad.flatten(), atol=1e-5)
     assert torch.allclose(grad[K].cpu(), bc.grad[0], atol=1e-5)
+
+
+def test_toy_fused_inkernel_sgd():
+    # lr > 0 path: one launch does fwd+bwd+SGD; matches torch step exactly
+    B, K, LR = 32, 20, 0.1
+    x = _rand(B, K, seed=40).to(DEV)
+    t = _rand(B, 1, seed=41).to(DEV)
+    w = (_rand(1, K, seed=42) - 0.5).to(DEV)
+    b = _rand(1, seed=43).to(DEV)
+    param = torch.zeros(24, device=DEV)
+    param[:K] = w.flatten()
+    param[K] = b[0]
+    grad = torch.zeros(24, device=DEV)
+    ops.ext().toy_fused_fwd_bwd(x, t, param, grad, torch.Tensor(), True,
+                                0, K, LR)
+    wc = w.cpu().requires_grad_(True)
+    bc = b.cpu().requires_grad_(True)
+    loss = torch.nn.functional.mse_loss(
+        torch.nn.functional.linear(x.cpu(), wc, bc), t.cpu())
+    loss.backward()
+    assert torch.allclose(param[:K].cpu(), (wc - LR * wc.grad).detach().flatten(),
+                          atol=1e-5)
+    assert torch.allclose(param[K].cpu(), (bc - LR * bc.grad).detach()[0],
+                          atol=1e-5)
+    assert grad.abs().sum() == 0  # grads never materialized
+
+
+def test_rccl_world1_comm():
+    # exercises the native RCCL path end-to-end at world 1 (uid creation,
+    # comm init, comm-stream all-reduce, event fencing, broadcast, barrier)
+    C = ops.ext()
+    uid = C.RcclComm.make_unique_id()
+    comm = C.RcclComm(uid, 0, 1, 0)
+    t = _rand(1024, seed=50).to(DEV)
+    ref = t.clone()
+    comm.all_reduce_avg(t)   # avg over world 1 == identity
+    comm.join_compute()
+    torch.cuda.synchronize()
+    assert torch.allclose(t, ref)
+    comm.all_reduce_avg_inline(t)
+    comm.broadcast(t, 0)
+    comm.barrier()
+    assert torch.allclose(t, ref)
+    del comm
 
 
 def test_single_gpu_trainer_runs(tmp_chdir):
