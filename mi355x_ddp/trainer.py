@@ -83,7 +83,13 @@ class Trainer:
             gpu_id = int(os.environ.get("LOCAL_RANK", 0))
         self.gpu_id = gpu_id
         self.device = torch.device("cpu") if gpu_id == "cpu" else torch.device("cuda", gpu_id)
-        self.global_rank = int(os.environ.get("RANK", 0))
+        # global rank: the process group is authoritative (mp.spawn sets no
+        # RANK env; torchrun does — reference multinode_torchrun.py:25)
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized():
+            self.global_rank = dist.get_rank()
+        else:
+            self.global_rank = int(os.environ.get("RANK", 0))
         self.train_data = train_data
         self.optimizer = optimizer
         self.save_every = save_every
@@ -156,7 +162,28 @@ class Trainer:
             self.model.finalize_backward()
         self.optimizer.step()
 
+    def _maybe_inject_fault(self, epoch: int) -> None:
+        """Fault injection for elastic-restart testing (the
+        reference has no injection tooling; resume-by-kill is its implied
+        validation). When MI355X_FAULT_EPOCH matches and the one-shot marker
+        file does not exist yet, this rank hard-exits — torchrun's elastic
+        agent then restarts the job, which must resume from the snapshot."""
+        if os.environ.get("MI355X_FAULT_EPOCH") != str(epoch):
+            return
+        if self.global_rank != int(os.environ.get("MI355X_FAULT_RANK", "0")):
+            return
+        marker = os.environ.get("MI355X_FAULT_ONCE_FILE")
+        if marker:
+            if os.path.exists(marker):
+                return  # already crashed once; run through this time
+            with open(marker, "w") as f:
+                f.write("crashed\n")
+        print(f"[GPU{self.global_rank}] injected fault at epoch {epoch}",
+              flush=True)
+        os._exit(17)
+
     def _run_epoch(self, epoch: int) -> None:
+        self._maybe_inject_fault(epoch)
         b_sz = self.train_data.batch_size
         print(f"[GPU{self.global_rank if self._distributed else self.gpu_id}] "
               f"Epoch {epoch} | Batchsize: {b_sz} | Steps: {len(self.train_data)}")

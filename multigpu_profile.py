@@ -26,9 +26,24 @@ from mi355x_ddp.parallel import FusedSGD, ddp_setup
 from mi355x_ddp.trainer import Trainer
 
 
+def _tiny_cnn():
+    # CPU test stand-in (MI355X_PROFILE_MODEL=tiny): keeps the multi-bucket
+    # reducer + profiler machinery exercised without ResNet-50's CPU cost.
+    import torch.nn as nn
+    return nn.Sequential(
+        nn.Conv2d(3, 8, 3, stride=2, padding=1), nn.BatchNorm2d(8),
+        nn.ReLU(), nn.AdaptiveAvgPool2d(1), nn.Flatten(),
+        nn.Linear(8, 1000))
+
+
 def load_train_objs(dataset_size: int = 2048):
-    train_set = RandomImageDataset(dataset_size, (3, 224, 224))
-    model = resnet50()
+    shape = (3, 224, 224)
+    if os.environ.get("MI355X_PROFILE_MODEL") == "tiny":
+        shape = (3, 32, 32)
+        model = _tiny_cnn()
+    else:
+        model = resnet50()
+    train_set = RandomImageDataset(dataset_size, shape)
     optimizer = FusedSGD(model.parameters(), lr=1e-3)
     return train_set, model, optimizer
 
@@ -56,8 +71,11 @@ def main(rank: int, world_size: int, total_epochs: int = 3,
 
 if __name__ == "__main__":
     total_epochs = int(sys.argv[1]) if len(sys.argv) > 1 else 3
+    dataset_size = int(os.environ.get("MI355X_PROFILE_DATASET", 2048))
+    batch_size = int(os.environ.get("MI355X_PROFILE_BATCH", 32))
     if torch.cuda.is_available():
         world_size = torch.cuda.device_count()
     else:
         world_size = int(os.environ.get("MI355X_WORLD", 2))
-    mp.spawn(main, args=(world_size, total_epochs), nprocs=world_size)
+    mp.spawn(main, args=(world_size, total_epochs, True, dataset_size,
+                         batch_size), nprocs=world_size)
