@@ -37,6 +37,8 @@ class HipLinear(nn.Module):
             nn.init.uniform_(self.bias, -bound, bound)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dtype != self.weight.dtype:
+            x = x.to(self.weight.dtype)  # bf16 models take f32 loader batches
         return ops.linear(x, self.weight, self.bias)
 
     def extra_repr(self) -> str:

@@ -15,6 +15,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("dy"), py::arg("dw"), py::arg("db"),
         py::arg("accumulate") = false);
   m.def("linear_bwd_input", &mi355x::linear_bwd_input);
+  m.def("gemm_bf16", &mi355x::gemm_bf16, py::arg("x"), py::arg("w"),
+        py::arg("bias") = c10::nullopt);
   m.def("ce_fwd", &mi355x::ce_fwd);
   m.def("ce_bwd", &mi355x::ce_bwd);
   m.def("mse_fwd", &mi355x::mse_fwd);

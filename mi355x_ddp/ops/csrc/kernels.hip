@@ -9,12 +9,16 @@
 // (single_gpu.py:26), N9 zero_grad folded into the bucket lifecycle
 // (single_gpu.py:22), and the reducer's bucket flatten/unflatten (N3).
 //
-// f32 matmuls use the exact-f32 MFMA `v_mfma_f32_16x16x4_f32`
-// (f32-in/f32-acc at the f32 vector rate): the toy shapes are single-wave
-// and latency-bound, so the MFMA path buys kernel-count and issue-slot
-// economy, not FLOPs — see SURVEY.md §7 step 2.
+// Dtypes: every kernel is templated on the STORAGE type (f32 or bf16;
+// bf16 is the BASELINE.json config-2 capability). Matmul compute uses
+// the exact-f32 MFMA `v_mfma_f32_16x16x4_f32` for f32 storage (no
+// xf32/TF32 exists on gfx950 — cdna_hip_programming.md §3) and
+// `v_mfma_f32_16x16x32_bf16` for bf16 storage in the standalone GEMM; the
+// toy shapes are single-wave and latency-bound, so the MFMA path buys
+// kernel-count and issue-slot economy, not FLOPs.
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 #include <c10/hip/HIPStream.h>
@@ -27,9 +31,45 @@
     TORCH_CHECK(_e == hipSuccess, "HIP error: ", hipGetErrorString(_e));      \
   } while (0)
 
+// dispatch on storage dtype (f32 | bf16)
+#define DISPATCH_F32_BF16(TYPE, NAME, ...)                                    \
+  switch (TYPE) {                                                             \
+    case at::kFloat: {                                                        \
+      using scalar_t = float;                                                 \
+      __VA_ARGS__;                                                            \
+      break;                                                                  \
+    }                                                                         \
+    case at::kBFloat16: {                                                     \
+      using scalar_t = __hip_bfloat16;                                        \
+      __VA_ARGS__;                                                            \
+      break;                                                                  \
+    }                                                                         \
+    default:                                                                  \
+      TORCH_CHECK(false, NAME, ": unsupported dtype ", TYPE);                 \
+  }
+
 namespace mi355x {
 
 using f32x4 = __attribute__((ext_vector_type(4))) float;
+using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ float ldf(const float* p) { return *p; }
+__device__ __forceinline__ float ldf(const __hip_bfloat16* p) {
+  return __bfloat162float(*p);
+}
+__device__ __forceinline__ void stf(float* p, float v) { *p = v; }
+__device__ __forceinline__ void stf(__hip_bfloat16* p, float v) {
+  *p = __float2bfloat16(v);
+}
+
+template <typename T>
+static T* dptr(torch::Tensor& t) {
+  return reinterpret_cast<T*>(t.data_ptr());
+}
+template <typename T>
+static const T* cdptr(const torch::Tensor& t) {
+  return reinterpret_cast<const T*>(t.data_ptr());
+}
 
 static inline hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
@@ -37,17 +77,28 @@ static inline hipStream_t cur_stream() {
 
 static inline int cdiv(int64_t a, int64_t b) { return (int)((a + b - 1) / b); }
 
+__device__ __forceinline__ float wave_max(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+__device__ __forceinline__ float wave_sum(float v) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
 // ---------------------------------------------------------------------------
 // Linear forward: Y[B,N] = X[B,K] @ W[N,K]^T + bias[N]
-// One wave (64 lanes) per 16x16 output tile; K-loop of mfma_f32_16x16x4f32.
+// One wave (64 lanes) per 16x16 output tile; K-loop of mfma_f32_16x16x4f32
+// (storage loads upconvert bf16 -> f32; exact-f32 accumulate).
 // Lane maps (cdna_hip_programming.md §3): A[l&15][k=l>>4], B[k=l>>4][l&15],
 // C/D col=lane&15, row=(lane>>4)*4+reg.
 // ---------------------------------------------------------------------------
-__global__ void k_linear_fwd_f32(const float* __restrict__ X,
-                                 const float* __restrict__ W,
-                                 const float* __restrict__ bias,
-                                 float* __restrict__ Y,
-                                 int B, int K, int N) {
+template <typename T>
+__global__ void k_linear_fwd(const T* __restrict__ X, const T* __restrict__ W,
+                             const T* __restrict__ bias, T* __restrict__ Y,
+                             int B, int K, int N) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;  // 4 waves/block, each one M-tile
   const int tile_m = blockIdx.x * 4 + wave;
@@ -60,37 +111,41 @@ __global__ void k_linear_fwd_f32(const float* __restrict__ X,
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   for (int k0 = 0; k0 < K; k0 += 4) {
     const int k = k0 + q;
-    const float a = (m < B && k < K) ? X[(size_t)m * K + k] : 0.f;
-    const float b = (n < N && k < K) ? W[(size_t)n * K + k] : 0.f;
+    const float a = (m < B && k < K) ? ldf(&X[(size_t)m * K + k]) : 0.f;
+    const float b = (n < N && k < K) ? ldf(&W[(size_t)n * K + k]) : 0.f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
   const int col = tile_n * 16 + r;
   if (col < N) {
-    const float bv = bias ? bias[col] : 0.f;
+    const float bv = bias ? ldf(&bias[col]) : 0.f;
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int row = tile_m * 16 + q * 4 + i;
-      if (row < B) Y[(size_t)row * N + col] = acc[i] + bv;
+      if (row < B) stf(&Y[(size_t)row * N + col], acc[i] + bv);
     }
   }
 }
 
 torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
                          c10::optional<torch::Tensor> bias) {
-  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kFloat, "x must be f32 on device");
+  TORCH_CHECK(x.is_cuda(), "x must be on device");
   TORCH_CHECK(x.dim() == 2 && w.dim() == 2 && x.size(1) == w.size(1),
               "shape mismatch for linear");
+  TORCH_CHECK(x.scalar_type() == w.scalar_type(), "x/w dtype mismatch");
   auto xc = x.contiguous();
   auto wc = w.contiguous();
   const int B = (int)xc.size(0), K = (int)xc.size(1), N = (int)wc.size(0);
   auto y = at::empty({B, N}, x.options());
-  const float* bp = nullptr;
   torch::Tensor bc;
-  if (bias.has_value()) { bc = bias->contiguous(); bp = bc.data_ptr<float>(); }
+  bool has_bias = bias.has_value();
+  if (has_bias) bc = bias->contiguous();
   dim3 grid(cdiv(B, 64), cdiv(N, 16));
-  hipLaunchKernelGGL(k_linear_fwd_f32, grid, dim3(256), 0, cur_stream(),
-                     xc.data_ptr<float>(), wc.data_ptr<float>(), bp,
-                     y.data_ptr<float>(), B, K, N);
+  DISPATCH_F32_BF16(x.scalar_type(), "linear_fwd", {
+    hipLaunchKernelGGL((k_linear_fwd<scalar_t>), grid, dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(xc), cdptr<scalar_t>(wc),
+                       has_bias ? cdptr<scalar_t>(bc) : nullptr,
+                       dptr<scalar_t>(y), B, K, N);
+  });
   HIP_OK(hipGetLastError());
   return y;
 }
@@ -100,11 +155,11 @@ torch::Tensor linear_fwd(torch::Tensor x, torch::Tensor w,
 // Output tile (16 rows of N) x (16 cols of K) per wave; contraction over B.
 // A[j][i] = dY[i][j]; B-operand[i][k] = X[i][k].
 // ---------------------------------------------------------------------------
-__global__ void k_linear_bwd_w_f32(const float* __restrict__ X,
-                                   const float* __restrict__ dY,
-                                   float* __restrict__ dW,
-                                   float* __restrict__ dB,
-                                   int B, int K, int N, int accumulate) {
+template <typename T>
+__global__ void k_linear_bwd_w(const T* __restrict__ X,
+                               const T* __restrict__ dY,
+                               T* __restrict__ dW, T* __restrict__ dB,
+                               int B, int K, int N, int accumulate) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int tile_k = blockIdx.x * 4 + wave;  // K-tile (output cols)
@@ -118,8 +173,8 @@ __global__ void k_linear_bwd_w_f32(const float* __restrict__ X,
   if (tile_k * 16 < K) {
     for (int i0 = 0; i0 < B; i0 += 4) {
       const int i = i0 + q;
-      const float a = (i < B && j < N) ? dY[(size_t)i * N + j] : 0.f;
-      const float b = (i < B && k < K) ? X[(size_t)i * K + k] : 0.f;
+      const float a = (i < B && j < N) ? ldf(&dY[(size_t)i * N + j]) : 0.f;
+      const float b = (i < B && k < K) ? ldf(&X[(size_t)i * K + k]) : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
       if (tile_k == 0) bsum += a;  // reuse the loaded dY for db
     }
@@ -129,15 +184,15 @@ __global__ void k_linear_bwd_w_f32(const float* __restrict__ X,
       for (int i = 0; i < 4; ++i) {
         const int row = tile_n * 16 + q * 4 + i;
         if (row < N) {
-          const size_t off = (size_t)row * K + col;
-          dW[off] = acc[i] + (accumulate ? dW[off] : 0.f);
+          T* out = &dW[(size_t)row * K + col];
+          stf(out, acc[i] + (accumulate ? ldf(out) : 0.f));
         }
       }
     }
   } else if (tile_k == 0) {
     for (int i0 = 0; i0 < B; i0 += 4) {
       const int i = i0 + q;
-      bsum += (i < B && j < N) ? dY[(size_t)i * N + j] : 0.f;
+      bsum += (i < B && j < N) ? ldf(&dY[(size_t)i * N + j]) : 0.f;
     }
   }
   if (tile_k == 0 && dB) {
@@ -145,7 +200,7 @@ __global__ void k_linear_bwd_w_f32(const float* __restrict__ X,
     bsum += __shfl_xor(bsum, 16, 64);
     bsum += __shfl_xor(bsum, 32, 64);
     if (q == 0 && j < N) {
-      if (accumulate) dB[j] += bsum; else dB[j] = bsum;
+      stf(&dB[j], bsum + (accumulate ? ldf(&dB[j]) : 0.f));
     }
   }
 }
@@ -156,19 +211,22 @@ void linear_bwd_weight(torch::Tensor x, torch::Tensor dy,
   auto dyc = dy.contiguous();
   TORCH_CHECK(dw.is_contiguous() && db.is_contiguous(), "grad views must be contiguous");
   const int B = (int)xc.size(0), K = (int)xc.size(1), N = (int)dyc.size(1);
-  dim3 grid(cdiv(K, 64) , cdiv(N, 16));
-  hipLaunchKernelGGL(k_linear_bwd_w_f32, grid, dim3(256), 0, cur_stream(),
-                     xc.data_ptr<float>(), dyc.data_ptr<float>(),
-                     dw.data_ptr<float>(), db.numel() ? db.data_ptr<float>() : nullptr,
-                     B, K, N, accumulate ? 1 : 0);
+  dim3 grid(cdiv(K, 64), cdiv(N, 16));
+  DISPATCH_F32_BF16(x.scalar_type(), "linear_bwd_weight", {
+    hipLaunchKernelGGL((k_linear_bwd_w<scalar_t>), grid, dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(xc), cdptr<scalar_t>(dyc),
+                       dptr<scalar_t>(dw),
+                       db.numel() ? dptr<scalar_t>(db) : nullptr,
+                       B, K, N, accumulate ? 1 : 0);
+  });
   HIP_OK(hipGetLastError());
 }
 
 // dX[B,K] = dY[B,N] @ W[N,K]
-__global__ void k_linear_bwd_x_f32(const float* __restrict__ dY,
-                                   const float* __restrict__ W,
-                                   float* __restrict__ dX,
-                                   int B, int K, int N) {
+template <typename T>
+__global__ void k_linear_bwd_x(const T* __restrict__ dY,
+                               const T* __restrict__ W, T* __restrict__ dX,
+                               int B, int K, int N) {
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int tile_m = blockIdx.x * 4 + wave;
@@ -181,8 +239,8 @@ __global__ void k_linear_bwd_x_f32(const float* __restrict__ dY,
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   for (int j0 = 0; j0 < N; j0 += 4) {
     const int j = j0 + q;
-    const float a = (m < B && j < N) ? dY[(size_t)m * N + j] : 0.f;
-    const float b = (j < N && kc < K) ? W[(size_t)j * K + kc] : 0.f;
+    const float a = (m < B && j < N) ? ldf(&dY[(size_t)m * N + j]) : 0.f;
+    const float b = (j < N && kc < K) ? ldf(&W[(size_t)j * K + kc]) : 0.f;
     acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
   const int col = tile_k * 16 + r;
@@ -190,7 +248,7 @@ __global__ void k_linear_bwd_x_f32(const float* __restrict__ dY,
 #pragma unroll
     for (int i = 0; i < 4; ++i) {
       const int row = tile_m * 16 + q * 4 + i;
-      if (row < B) dX[(size_t)row * K + col] = acc[i];
+      if (row < B) stf(&dX[(size_t)row * K + col], acc[i]);
     }
   }
 }
@@ -201,53 +259,114 @@ torch::Tensor linear_bwd_input(torch::Tensor dy, torch::Tensor w) {
   const int B = (int)dyc.size(0), N = (int)dyc.size(1), K = (int)wc.size(1);
   auto dx = at::empty({B, K}, dy.options());
   dim3 grid(cdiv(B, 64), cdiv(K, 16));
-  hipLaunchKernelGGL(k_linear_bwd_x_f32, grid, dim3(256), 0, cur_stream(),
-                     dyc.data_ptr<float>(), wc.data_ptr<float>(),
-                     dx.data_ptr<float>(), B, K, N);
+  DISPATCH_F32_BF16(dy.scalar_type(), "linear_bwd_input", {
+    hipLaunchKernelGGL((k_linear_bwd_x<scalar_t>), grid, dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(dyc), cdptr<scalar_t>(wc),
+                       dptr<scalar_t>(dx), B, K, N);
+  });
   HIP_OK(hipGetLastError());
   return dx;
+}
+
+// ---------------------------------------------------------------------------
+// Standalone bf16 MFMA GEMM: Y[B,N] = X[B,K] @ W[N,K]^T (+bias), f32
+// accumulate via v_mfma_f32_16x16x32_bf16 (the gfx950 2xK form). One wave
+// per 16x16 tile, K consumed 32 at a time, 8 bf16 per lane per operand.
+// A: lane l holds A[l&15][(l>>4)*8 + j], j=0..7; B-op: W[l&15][(l>>4)*8+j];
+// C/D: col=lane&15, row=(lane>>4)*4+reg (cdna_hip_programming.md §3).
+// Used for larger HipLinear shapes in bf16 models; verified against torch
+// in tests/test_kernels_gpu.py.
+// ---------------------------------------------------------------------------
+__global__ void k_gemm_bf16(const __hip_bfloat16* __restrict__ X,
+                            const __hip_bfloat16* __restrict__ W,
+                            const __hip_bfloat16* __restrict__ bias,
+                            __hip_bfloat16* __restrict__ Y,
+                            int B, int K, int N) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int tile_m = blockIdx.x * 4 + wave;
+  const int tile_n = blockIdx.y;
+  if (tile_m * 16 >= B) return;
+  const int r = lane & 15;
+  const int q = lane >> 4;
+  const int m = tile_m * 16 + r;
+  const int n = tile_n * 16 + r;
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  for (int k0 = 0; k0 < K; k0 += 32) {
+    bf16x8 a{}, b{};
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = k0 + q * 8 + j;
+      a[j] = (m < B && k < K)
+                 ? *reinterpret_cast<const __bf16*>(&X[(size_t)m * K + k])
+                 : (__bf16)0.f;
+      b[j] = (n < N && k < K)
+                 ? *reinterpret_cast<const __bf16*>(&W[(size_t)n * K + k])
+                 : (__bf16)0.f;
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
+  }
+  const int col = tile_n * 16 + r;
+  if (col < N) {
+    const float bv = bias ? ldf(&bias[col]) : 0.f;
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int row = tile_m * 16 + q * 4 + i;
+      if (row < B) stf(&Y[(size_t)row * N + col], acc[i] + bv);
+    }
+  }
+}
+
+torch::Tensor gemm_bf16(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias) {
+  TORCH_CHECK(x.is_cuda() && x.scalar_type() == at::kBFloat16);
+  auto xc = x.contiguous();
+  auto wc = w.contiguous();
+  const int B = (int)xc.size(0), K = (int)xc.size(1), N = (int)wc.size(0);
+  auto y = at::empty({B, N}, x.options());
+  torch::Tensor bc;
+  bool has_bias = bias.has_value();
+  if (has_bias) bc = bias->contiguous();
+  dim3 grid(cdiv(B, 64), cdiv(N, 16));
+  hipLaunchKernelGGL(k_gemm_bf16, grid, dim3(256), 0, cur_stream(),
+                     cdptr<__hip_bfloat16>(xc), cdptr<__hip_bfloat16>(wc),
+                     has_bias ? cdptr<__hip_bfloat16>(bc) : nullptr,
+                     dptr<__hip_bfloat16>(y), B, K, N);
+  HIP_OK(hipGetLastError());
+  return y;
 }
 
 // ---------------------------------------------------------------------------
 // Cross-entropy with probability targets (torch semantics:
 // loss = mean_i [ -sum_c t_ic * log_softmax(y_i)_c ]).
 // One wave per row; lanes stride over C; wave shuffle reductions (wave=64).
-// Saves probs and per-row target sums for the backward.
+// probs/tsum/loss are saved in f32 regardless of storage dtype.
 // ---------------------------------------------------------------------------
-__device__ __forceinline__ float wave_max(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
-  return v;
-}
-__device__ __forceinline__ float wave_sum(float v) {
-#pragma unroll
-  for (int off = 32; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
-  return v;
-}
-
-__global__ void k_ce_fwd_f32(const float* __restrict__ Y,
-                             const float* __restrict__ T,
-                             float* __restrict__ P,
-                             float* __restrict__ tsum,
-                             float* __restrict__ loss,  // pre-zeroed scalar
-                             int B, int C) {
+template <typename T>
+__global__ void k_ce_fwd(const T* __restrict__ Y, const T* __restrict__ Tg,
+                         float* __restrict__ P, float* __restrict__ tsum,
+                         float* __restrict__ loss,  // pre-zeroed scalar
+                         int B, int C) {
   const int lane = threadIdx.x & 63;
   const int row = blockIdx.x * (blockDim.x >> 6) + (threadIdx.x >> 6);
   if (row >= B) return;
-  const float* y = Y + (size_t)row * C;
-  const float* t = T + (size_t)row * C;
+  const T* y = Y + (size_t)row * C;
+  const T* t = Tg + (size_t)row * C;
   float* p = P + (size_t)row * C;
   float m = -INFINITY;
-  for (int c = lane; c < C; c += 64) m = fmaxf(m, y[c]);
+  for (int c = lane; c < C; c += 64) m = fmaxf(m, ldf(&y[c]));
   m = wave_max(m);
   float se = 0.f, ts = 0.f, ty = 0.f;
   for (int c = lane; c < C; c += 64) {
-    const float e = __expf(y[c] - m);
-    se += e; ts += t[c]; ty += t[c] * y[c];
+    const float yv = ldf(&y[c]);
+    const float tv = ldf(&t[c]);
+    se += __expf(yv - m);
+    ts += tv;
+    ty += tv * yv;
   }
   se = wave_sum(se); ts = wave_sum(ts); ty = wave_sum(ty);
   const float inv_se = 1.f / se;
-  for (int c = lane; c < C; c += 64) p[c] = __expf(y[c] - m) * inv_se;
+  for (int c = lane; c < C; c += 64) p[c] = __expf(ldf(&y[c]) - m) * inv_se;
   if (lane == 0) {
     tsum[row] = ts;
     const float logZ = m + __logf(se);
@@ -259,40 +378,45 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor y, torch::Tensor t) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
   const int B = (int)yc.size(0), C = (int)yc.size(1);
-  auto probs = at::empty_like(yc);
-  auto tsum = at::empty({B}, yc.options());
-  auto loss = at::zeros({}, yc.options());
+  auto f32opt = yc.options().dtype(at::kFloat);
+  auto probs = at::empty({B, C}, f32opt);
+  auto tsum = at::empty({B}, f32opt);
+  auto loss = at::zeros({}, f32opt);
   dim3 grid(cdiv(B, 4));
-  hipLaunchKernelGGL(k_ce_fwd_f32, grid, dim3(256), 0, cur_stream(),
-                     yc.data_ptr<float>(), tc.data_ptr<float>(),
-                     probs.data_ptr<float>(), tsum.data_ptr<float>(),
-                     loss.data_ptr<float>(), B, C);
+  DISPATCH_F32_BF16(y.scalar_type(), "ce_fwd", {
+    hipLaunchKernelGGL((k_ce_fwd<scalar_t>), grid, dim3(256), 0, cur_stream(),
+                       cdptr<scalar_t>(yc), cdptr<scalar_t>(tc),
+                       probs.data_ptr<float>(), tsum.data_ptr<float>(),
+                       loss.data_ptr<float>(), B, C);
+  });
   HIP_OK(hipGetLastError());
   return {loss, probs, tsum};
 }
 
 // dY = (tsum_row * p - t) * grad_scale / B
-__global__ void k_ce_bwd_f32(const float* __restrict__ P,
-                             const float* __restrict__ T,
-                             const float* __restrict__ tsum,
-                             float* __restrict__ dY,
-                             float scale, int B, int C) {
+template <typename T>
+__global__ void k_ce_bwd(const float* __restrict__ P, const T* __restrict__ Tg,
+                         const float* __restrict__ tsum, T* __restrict__ dY,
+                         float scale, int B, int C) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t n = (int64_t)B * C;
   if (i >= n) return;
   const int row = (int)(i / C);
-  dY[i] = (tsum[row] * P[i] - T[i]) * scale;
+  stf(&dY[i], (tsum[row] * P[i] - ldf(&Tg[i])) * scale);
 }
 
 torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor t,
                      torch::Tensor tsum, double grad_scale) {
   const int B = (int)probs.size(0), C = (int)probs.size(1);
-  auto dy = at::empty_like(probs);
+  auto tc = t.contiguous();
+  auto dy = at::empty({B, C}, tc.options());
   const int64_t n = (int64_t)B * C;
-  hipLaunchKernelGGL(k_ce_bwd_f32, dim3(cdiv(n, 256)), dim3(256), 0, cur_stream(),
-                     probs.data_ptr<float>(), t.contiguous().data_ptr<float>(),
-                     tsum.data_ptr<float>(), dy.data_ptr<float>(),
-                     (float)(grad_scale / B), B, C);
+  DISPATCH_F32_BF16(tc.scalar_type(), "ce_bwd", {
+    hipLaunchKernelGGL((k_ce_bwd<scalar_t>), dim3(cdiv(n, 256)), dim3(256), 0,
+                       cur_stream(), probs.data_ptr<float>(),
+                       cdptr<scalar_t>(tc), tsum.data_ptr<float>(),
+                       dptr<scalar_t>(dy), (float)(grad_scale / B), B, C);
+  });
   HIP_OK(hipGetLastError());
   return dy;
 }
@@ -300,14 +424,14 @@ torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor t,
 // ---------------------------------------------------------------------------
 // MSE: loss = mean((y - t)^2); dY = 2 (y - t) * grad_scale / numel
 // ---------------------------------------------------------------------------
-__global__ void k_mse_fwd_f32(const float* __restrict__ Y,
-                              const float* __restrict__ T,
-                              float* __restrict__ loss,  // pre-zeroed
-                              int64_t n, float inv_n) {
+template <typename T>
+__global__ void k_mse_fwd(const T* __restrict__ Y, const T* __restrict__ Tg,
+                          float* __restrict__ loss,  // pre-zeroed
+                          int64_t n, float inv_n) {
   float acc = 0.f;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
-    const float d = Y[i] - T[i];
+    const float d = ldf(&Y[i]) - ldf(&Tg[i]);
     acc += d * d;
   }
   acc = wave_sum(acc);
@@ -326,20 +450,22 @@ torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
   const int64_t n = yc.numel();
-  auto loss = at::zeros({}, yc.options());
+  auto loss = at::zeros({}, yc.options().dtype(at::kFloat));
   const int blocks = (int)std::min<int64_t>(cdiv(n, 256), 2048);
-  hipLaunchKernelGGL(k_mse_fwd_f32, dim3(blocks), dim3(256), 0, cur_stream(),
-                     yc.data_ptr<float>(), tc.data_ptr<float>(),
-                     loss.data_ptr<float>(), n, 1.f / (float)n);
+  DISPATCH_F32_BF16(y.scalar_type(), "mse_fwd", {
+    hipLaunchKernelGGL((k_mse_fwd<scalar_t>), dim3(blocks), dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(yc), cdptr<scalar_t>(tc),
+                       loss.data_ptr<float>(), n, 1.f / (float)n);
+  });
   HIP_OK(hipGetLastError());
   return loss;
 }
 
-__global__ void k_mse_bwd_f32(const float* __restrict__ Y,
-                              const float* __restrict__ T,
-                              float* __restrict__ dY, float scale, int64_t n) {
+template <typename T>
+__global__ void k_mse_bwd(const T* __restrict__ Y, const T* __restrict__ Tg,
+                          T* __restrict__ dY, float scale, int64_t n) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) dY[i] = (Y[i] - T[i]) * scale;
+  if (i < n) stf(&dY[i], (ldf(&Y[i]) - ldf(&Tg[i])) * scale);
 }
 
 torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale) {
@@ -347,17 +473,19 @@ torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale) {
   auto tc = t.contiguous();
   const int64_t n = yc.numel();
   auto dy = at::empty_like(yc);
-  hipLaunchKernelGGL(k_mse_bwd_f32, dim3(cdiv(n, 256)), dim3(256), 0, cur_stream(),
-                     yc.data_ptr<float>(), tc.data_ptr<float>(),
-                     dy.data_ptr<float>(), (float)(2.0 * grad_scale / n), n);
+  DISPATCH_F32_BF16(y.scalar_type(), "mse_bwd", {
+    hipLaunchKernelGGL((k_mse_bwd<scalar_t>), dim3(cdiv(n, 256)), dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(yc), cdptr<scalar_t>(tc),
+                       dptr<scalar_t>(dy), (float)(2.0 * grad_scale / n), n);
+  });
   HIP_OK(hipGetLastError());
   return dy;
 }
 
 // ---------------------------------------------------------------------------
 // Fused SGD over a flat bucket (+ fold zero_grad): p -= lr * g; g = 0.
-// Vectorized float4 (16 B/lane); bucket lengths are padded to a multiple of
-// 4 elements by the reducer, so the f32x4 path covers everything.
+// 4 elements per lane (16 B for f32, 8 B for bf16); bucket lengths are
+// padded to a multiple of 4 elements by the reducer.
 // ---------------------------------------------------------------------------
 __global__ void k_sgd_flat_f32(float4* __restrict__ p, float4* __restrict__ g,
                                float lr, int64_t n4, int zero) {
@@ -370,6 +498,27 @@ __global__ void k_sgd_flat_f32(float4* __restrict__ p, float4* __restrict__ g,
   if (zero) g[i] = make_float4(0.f, 0.f, 0.f, 0.f);
 }
 
+// bf16 variant: 4 elements = one 8-byte ushort4 per lane (vectorized load;
+// hipcc does not auto-vectorize scalar bf16 — guide G13)
+__global__ void k_sgd_flat_bf16(ushort4* __restrict__ p, ushort4* __restrict__ g,
+                                float lr, int64_t n4, int zero) {
+  const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n4) return;
+  ushort4 gv = g[i];
+  ushort4 pv = p[i];
+  unsigned short* gs = &gv.x;
+  unsigned short* ps = &pv.x;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const float pf = __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(ps + j));
+    const float gf = __bfloat162float(*reinterpret_cast<__hip_bfloat16*>(gs + j));
+    const __hip_bfloat16 out = __float2bfloat16(pf - lr * gf);
+    ps[j] = *reinterpret_cast<const unsigned short*>(&out);
+  }
+  p[i] = pv;
+  if (zero) g[i] = make_ushort4(0, 0, 0, 0);
+}
+
 void sgd_flat(torch::Tensor param_flat, torch::Tensor grad_flat,
               double lr, bool zero_grad) {
   TORCH_CHECK(param_flat.is_cuda() && param_flat.is_contiguous() &&
@@ -377,45 +526,53 @@ void sgd_flat(torch::Tensor param_flat, torch::Tensor grad_flat,
   TORCH_CHECK(param_flat.numel() % 4 == 0,
               "bucket length must be padded to a multiple of 4");
   const int64_t n4 = param_flat.numel() / 4;
-  hipLaunchKernelGGL(k_sgd_flat_f32, dim3(cdiv(n4, 256)), dim3(256), 0,
-                     cur_stream(),
-                     reinterpret_cast<float4*>(param_flat.data_ptr<float>()),
-                     reinterpret_cast<float4*>(grad_flat.data_ptr<float>()),
-                     (float)lr, n4, zero_grad ? 1 : 0);
+  const dim3 grid(cdiv(n4, 256));
+  if (param_flat.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL(k_sgd_flat_f32, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<float4*>(param_flat.data_ptr()),
+                       reinterpret_cast<float4*>(grad_flat.data_ptr()),
+                       (float)lr, n4, zero_grad ? 1 : 0);
+  } else if (param_flat.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL(k_sgd_flat_bf16, grid, dim3(256), 0, cur_stream(),
+                       reinterpret_cast<ushort4*>(param_flat.data_ptr()),
+                       reinterpret_cast<ushort4*>(grad_flat.data_ptr()),
+                       (float)lr, n4, zero_grad ? 1 : 0);
+  } else {
+    TORCH_CHECK(false, "sgd_flat: unsupported dtype");
+  }
   HIP_OK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
 // Bucket flatten/unflatten: one launch per bucket moves all
-// member tensors via a precomputed per-block copy plan.
-// plan row (int64 x3): [src_addr, bucket_elem_offset, numel] for each block's
-// chunk of ELEMS_PER_BLOCK elements. Tensor bucket offsets are 4-elem
-// aligned (reducer), torch allocations are 256-B aligned, so float4 moves
-// cover whole chunks; the (rare) tail of each tensor goes scalar.
+// member tensors <-> their byte ranges of the flat bucket via a precomputed
+// per-block copy plan. plan row (int64 x3): [src_addr_bytes,
+// bucket_byte_offset, nbytes]. Offsets are 4-element aligned and torch
+// allocations are 256-B aligned, so 8-byte vector moves cover whole chunks
+// for both f32 and bf16; tails go bytewise.
 // ---------------------------------------------------------------------------
-static constexpr int64_t ELEMS_PER_BLOCK = 2048;  // 256 threads x 8 f32
+static constexpr int64_t BYTES_PER_BLOCK = 8192;
 
 template <bool INTO_BUCKET>
-__global__ void k_bucket_copy(float* __restrict__ bucket,
+__global__ void k_bucket_copy(char* __restrict__ bucket,
                               const int64_t* __restrict__ plan,
                               int zero_src) {
   const int64_t* row = plan + (int64_t)blockIdx.x * 3;
-  float* tp = reinterpret_cast<float*>(row[0]);  // tensor chunk
-  float* bp = bucket + row[1];                   // bucket segment
+  char* tp = reinterpret_cast<char*>(row[0]);  // tensor chunk
+  char* bp = bucket + row[1];                  // bucket segment
   const int64_t n = row[2];
-  float* src = INTO_BUCKET ? tp : bp;
-  float* dst = INTO_BUCKET ? bp : tp;
-  const int64_t n4 = n >> 2;
-  float4* s4 = reinterpret_cast<float4*>(src);
-  float4* d4 = reinterpret_cast<float4*>(dst);
-  const float4 z4 = make_float4(0.f, 0.f, 0.f, 0.f);
-  for (int64_t i = threadIdx.x; i < n4; i += blockDim.x) {
-    d4[i] = s4[i];
-    if (zero_src) s4[i] = z4;  // fold zero_grad into the gather
+  char* src = INTO_BUCKET ? tp : bp;
+  char* dst = INTO_BUCKET ? bp : tp;
+  const int64_t n8 = n >> 3;
+  uint64_t* s8 = reinterpret_cast<uint64_t*>(src);
+  uint64_t* d8 = reinterpret_cast<uint64_t*>(dst);
+  for (int64_t i = threadIdx.x; i < n8; i += blockDim.x) {
+    d8[i] = s8[i];
+    if (zero_src) s8[i] = 0;  // fold zero_grad into the gather
   }
-  for (int64_t i = (n4 << 2) + threadIdx.x; i < n; i += blockDim.x) {
+  for (int64_t i = (n8 << 3) + threadIdx.x; i < n; i += blockDim.x) {
     dst[i] = src[i];
-    if (zero_src) src[i] = 0.f;
+    if (zero_src) src[i] = 0;
   }
 }
 
@@ -425,12 +582,14 @@ torch::Tensor build_copy_plan(const std::vector<torch::Tensor>& tensors,
   std::vector<int64_t> rows;
   for (size_t t = 0; t < tensors.size(); ++t) {
     TORCH_CHECK(tensors[t].is_contiguous(), "bucket members must be contiguous");
-    const int64_t numel = tensors[t].numel();
+    const int64_t esz = tensors[t].element_size();
+    const int64_t nbytes = tensors[t].numel() * esz;
     const int64_t addr = (int64_t)(uintptr_t)tensors[t].data_ptr();
-    for (int64_t base = 0; base < numel; base += ELEMS_PER_BLOCK) {
-      rows.push_back(addr + base * (int64_t)sizeof(float));
-      rows.push_back(offsets[t] + base);
-      rows.push_back(std::min(ELEMS_PER_BLOCK, numel - base));
+    const int64_t byte_off = offsets[t] * esz;
+    for (int64_t base = 0; base < nbytes; base += BYTES_PER_BLOCK) {
+      rows.push_back(addr + base);
+      rows.push_back(byte_off + base);
+      rows.push_back(std::min(BYTES_PER_BLOCK, nbytes - base));
     }
   }
   auto plan = torch::from_blob(rows.data(), {(int64_t)rows.size() / 3, 3},
@@ -442,52 +601,49 @@ void flatten_into(torch::Tensor bucket, torch::Tensor plan,
                   int64_t total_blocks, bool zero_src) {
   hipLaunchKernelGGL((k_bucket_copy<true>), dim3((uint32_t)total_blocks),
                      dim3(256), 0, cur_stream(),
-                     bucket.data_ptr<float>(), plan.data_ptr<int64_t>(),
-                     zero_src ? 1 : 0);
+                     reinterpret_cast<char*>(bucket.data_ptr()),
+                     plan.data_ptr<int64_t>(), zero_src ? 1 : 0);
   HIP_OK(hipGetLastError());
 }
 
 void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_blocks) {
   hipLaunchKernelGGL((k_bucket_copy<false>), dim3((uint32_t)total_blocks),
                      dim3(256), 0, cur_stream(),
-                     bucket.data_ptr<float>(), plan.data_ptr<int64_t>(), 0);
+                     reinterpret_cast<char*>(bucket.data_ptr()),
+                     plan.data_ptr<int64_t>(), 0);
   HIP_OK(hipGetLastError());
 }
 
 // ---------------------------------------------------------------------------
-// Fused toy training step (fwd + loss-grad + bwd in ONE kernel): the
-// reference hot loop single_gpu.py:21-26 for model = Linear(K,1), minus the
-// optimizer (which needs the all-reduced grads). Single workgroup, 64
-// threads (one wave): at 84 B of gradients the step is launch-latency
-// bound, so one launch replaces five (SURVEY §7 hard-part 2).
-// Layout: param_flat = [w(K) | b | pad], grad_flat same.
+// Fused toy training step (fwd + loss-grad + bwd [+ SGD] in ONE kernel):
+// the reference hot loop single_gpu.py:21-26 for model = Linear(K,1).
+// Single workgroup, 64 threads (one wave): at 84 B of gradients the step is
+// launch-latency bound, so one launch replaces five (SURVEY §7 hard-part 2).
+// lr > 0: apply SGD in-kernel (world-1 path — no all-reduce exists);
+// lr <= 0: write grads into the bucket for the all-reduce + sgd_flat path.
 // Supports B <= 128, K <= 32.
 // ---------------------------------------------------------------------------
-__global__ void k_toy_fused_f32(const float* __restrict__ X,
-                                const float* __restrict__ T,
-                                float* __restrict__ param,
-                                float* __restrict__ grad,
-                                float* __restrict__ loss_out,
-                                int B, int K, int use_mse,
-                                int w_off, int b_off, float lr) {
+template <typename T>
+__global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
+                            T* __restrict__ param, T* __restrict__ grad,
+                            float* __restrict__ loss_out,
+                            int B, int K, int use_mse,
+                            int w_off, int b_off, float lr) {
   const int lane = threadIdx.x;
   const int r = lane & 15, q = lane >> 4;
   __shared__ float dy_s[128];
 
   // forward: y_i = sum_k X[i,k] w_k + b ; MFMA tiles of 16 rows, j=0 column.
-  // w/b live at arbitrary offsets inside the reducer's flat bucket
-  // (reverse-registration order puts bias first).
-  const float bterm = param[b_off];
+  const float bterm = ldf(&param[b_off]);
   const int ntile = (B + 15) / 16;
-  float dy_own = 0.f;  // this lane's dY rows (written via LDS below)
   float loss_acc = 0.f;
   for (int tm = 0; tm < ntile; ++tm) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     const int m = tm * 16 + r;
     for (int k0 = 0; k0 < K; k0 += 4) {
       const int k = k0 + q;
-      const float a = (m < B && k < K) ? X[(size_t)m * K + k] : 0.f;
-      const float b = (r == 0 && k < K) ? param[w_off + k] : 0.f;  // B[k][j=0]
+      const float a = (m < B && k < K) ? ldf(&X[(size_t)m * K + k]) : 0.f;
+      const float b = (r == 0 && k < K) ? ldf(&param[w_off + k]) : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
     if (r == 0) {  // lanes 0,16,32,48 hold col j=0; rows q*4+i
@@ -496,7 +652,7 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
         const int row = tm * 16 + q * 4 + i;
         if (row < B) {
           const float y = acc[i] + bterm;
-          const float t = T[row];
+          const float t = ldf(&Tg[row]);
           float dy;
           if (use_mse) {
             const float d = y - t;
@@ -522,15 +678,13 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
     for (int i0 = 0; i0 < B; i0 += 4) {
       const int i = i0 + q;
       const float a = (r == 0 && i < B) ? dy_s[i] : 0.f;  // A[j=0][i]
-      const float b = (i < B && k < K) ? X[(size_t)i * K + k] : 0.f;
+      const float b = (i < B && k < K) ? ldf(&X[(size_t)i * K + k]) : 0.f;
       acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
     }
-    // D row j=0 lives in reg 0 of lanes with q==0; col = k.
-    // lr > 0: world-size-1 fast path — apply SGD in-kernel (no all-reduce
-    // needed, grads never materialize; one launch = one training step).
+    // D row j=0 lives in reg 0 of lanes with q==0; col = k
     if (q == 0 && k < K) {
-      if (lr > 0.f) param[w_off + k] -= lr * acc[0];
-      else grad[w_off + k] = acc[0];
+      if (lr > 0.f) stf(&param[w_off + k], ldf(&param[w_off + k]) - lr * acc[0]);
+      else stf(&grad[w_off + k], acc[0]);
     }
   }
   // db + loss reduce
@@ -539,8 +693,8 @@ __global__ void k_toy_fused_f32(const float* __restrict__ X,
   dbp = wave_sum(dbp);
   if (use_mse) loss_acc = wave_sum(loss_acc);
   if (lane == 0) {
-    if (lr > 0.f) param[b_off] -= lr * dbp;
-    else grad[b_off] = dbp;
+    if (lr > 0.f) stf(&param[b_off], bterm - lr * dbp);
+    else stf(&grad[b_off], dbp);
     if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
   }
 }
@@ -552,11 +706,18 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
   const int B = (int)x.size(0), K = (int)x.size(1);
   TORCH_CHECK(B <= 128 && K <= 32, "toy fused kernel supports B<=128, K<=32");
   TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
-  hipLaunchKernelGGL(k_toy_fused_f32, dim3(1), dim3(64), 0, cur_stream(),
-                     x.data_ptr<float>(), t.data_ptr<float>(),
-                     param_flat.data_ptr<float>(), grad_flat.data_ptr<float>(),
-                     loss_out.defined() ? loss_out.data_ptr<float>() : nullptr,
-                     B, K, use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr);
+  float* lossp = nullptr;
+  if (loss_out.defined() && loss_out.numel()) {
+    TORCH_CHECK(loss_out.scalar_type() == at::kFloat, "loss_out must be f32");
+    lossp = loss_out.data_ptr<float>();
+  }
+  DISPATCH_F32_BF16(x.scalar_type(), "toy_fused_fwd_bwd", {
+    hipLaunchKernelGGL((k_toy_fused<scalar_t>), dim3(1), dim3(64), 0,
+                       cur_stream(), cdptr<scalar_t>(x), cdptr<scalar_t>(t),
+                       dptr<scalar_t>(param_flat), dptr<scalar_t>(grad_flat),
+                       lossp, B, K, use_mse ? 1 : 0,
+                       (int)w_off, (int)b_off, (float)lr);
+  });
   HIP_OK(hipGetLastError());
 }
 

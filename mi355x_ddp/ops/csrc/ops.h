@@ -19,6 +19,11 @@ void linear_bwd_weight(torch::Tensor x, torch::Tensor dy,
 // dx = dy @ w
 torch::Tensor linear_bwd_input(torch::Tensor dy, torch::Tensor w);
 
+// Standalone bf16 MFMA GEMM (v_mfma_f32_16x16x32_bf16, f32 accumulate):
+// y = x @ w^T + bias for larger bf16 shapes.
+torch::Tensor gemm_bf16(torch::Tensor x, torch::Tensor w,
+                        c10::optional<torch::Tensor> bias);
+
 // Cross-entropy with class-probability targets (C may be 1: the reference's
 // degenerate Linear(20,1) case, single_gpu.py:24 + utils.py:7).
 // Returns (loss[scalar], probs[B,C], tsum[B]).
