@@ -71,7 +71,8 @@ def test_torchrun_elastic_restart_after_crash(tmp_path):
              cwd=tmp_path,
              env_extra={"MI355X_FAULT_EPOCH": "1",
                         "MI355X_FAULT_RANK": "1",
-                        "MI355X_FAULT_ONCE_FILE": str(marker)})
+                        "MI355X_FAULT_ONCE_FILE": str(marker)},
+             timeout=600)
     assert r.returncode == 0, (r.stdout[-1500:], r.stderr[-1500:])
     assert marker.exists()  # the crash really happened
     assert "injected fault at epoch 1" in r.stdout
