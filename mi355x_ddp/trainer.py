@@ -112,7 +112,9 @@ class Trainer:
 
         self._distributed = False
         import torch.distributed as dist
-        if wrap_ddp and dist.is_available() and dist.is_initialized() and dist.get_world_size() > 1:
+        if wrap_ddp and dist.is_available() and dist.is_initialized():
+            # world 1 wraps too: the engine still provides flat buckets and
+            # the fused SGD path (there is simply no communicator).
             from .parallel import DDP, FusedSGD
             self.model = DDP(self.model, bucket_cap_mb=bucket_cap_mb)
             self._distributed = True
