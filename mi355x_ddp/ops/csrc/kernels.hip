@@ -623,7 +623,14 @@ void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_bloc
 // lr <= 0: write grads into the bucket for the all-reduce + sgd_flat path.
 // Supports B <= 128, K <= 32.
 // ---------------------------------------------------------------------------
-template <typename T>
+// MT = M-tiles (ceil(B/16), compile-time), KT = K-tiles (ceil(K/16)).
+// Structure: one coalesced LDS burst loads X, t, w, bias up front (one
+// global round trip instead of per-MFMA dependent loads), then both MFMA
+// phases run off LDS with the tile accumulators interleaved so the 40-cycle
+// dependent-accumulator latency of v_mfma_f32_16x16x4_f32 hides behind the
+// other tile's issue (cdna_hip_programming.md §3: >=2 independent
+// accumulators reach the issue rate).
+template <typename T, int MT, int KT>
 __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
                             T* __restrict__ param, T* __restrict__ grad,
                             float* __restrict__ loss_out,
@@ -631,60 +638,86 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
                             int w_off, int b_off, float lr) {
   const int lane = threadIdx.x;
   const int r = lane & 15, q = lane >> 4;
+  __shared__ float xs[128 * 32];   // X staged [B][K]
+  __shared__ float ts[128];        // targets
+  __shared__ float ws[33];         // w (K) + bias at ws[32]
   __shared__ float dy_s[128];
 
-  // forward: y_i = sum_k X[i,k] w_k + b ; MFMA tiles of 16 rows, j=0 column.
-  const float bterm = ldf(&param[b_off]);
-  const int ntile = (B + 15) / 16;
+  {  // one coalesced staging burst (independent loads, one latency trip)
+    const int total = B * K;
+    for (int i = lane; i < total; i += 64) xs[i] = ldf(&X[i]);
+    for (int i = lane; i < B; i += 64) ts[i] = ldf(&Tg[i]);
+    if (lane < K) ws[lane] = ldf(&param[w_off + lane]);
+    if (lane == K) ws[32] = ldf(&param[b_off]);
+  }
+  __syncthreads();
+
+  // forward: y = X @ w + b; MT interleaved 16-row tiles, j=0 column only
+  const float bterm = ws[32];
   float loss_acc = 0.f;
-  for (int tm = 0; tm < ntile; ++tm) {
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const int m = tm * 16 + r;
+  {
+    f32x4 acc[MT];
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) acc[tm] = {0.f, 0.f, 0.f, 0.f};
     for (int k0 = 0; k0 < K; k0 += 4) {
       const int k = k0 + q;
-      const float a = (m < B && k < K) ? ldf(&X[(size_t)m * K + k]) : 0.f;
-      const float b = (r == 0 && k < K) ? ldf(&param[w_off + k]) : 0.f;
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
-    }
-    if (r == 0) {  // lanes 0,16,32,48 hold col j=0; rows q*4+i
+      const float b = (r == 0 && k < K) ? ws[k] : 0.f;  // B[k][j=0]
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row = tm * 16 + q * 4 + i;
-        if (row < B) {
-          const float y = acc[i] + bterm;
-          const float t = ldf(&Tg[row]);
-          float dy;
-          if (use_mse) {
-            const float d = y - t;
-            loss_acc += d * d;
-            dy = 2.f * d / (float)B;
-          } else {
-            // CE over one logit: log_softmax == 0 -> loss == 0, dY == 0
-            // (the reference's degenerate loss).
-            dy = 0.f;
+      for (int tm = 0; tm < MT; ++tm) {
+        const int m = tm * 16 + r;
+        const float a = (m < B && k < K) ? xs[m * K + k] : 0.f;
+        acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tm], 0, 0, 0);
+      }
+    }
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) {
+      if (r == 0) {  // lanes 0,16,32,48 hold col j=0; rows q*4+i
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int row = tm * 16 + q * 4 + i;
+          if (row < B) {
+            const float y = acc[tm][i] + bterm;
+            float dy;
+            if (use_mse) {
+              const float d = y - ts[row];
+              loss_acc += d * d;
+              dy = 2.f * d / (float)B;
+            } else {
+              // CE over one logit: log_softmax == 0 -> loss == 0, dY == 0
+              // (the reference's degenerate loss).
+              dy = 0.f;
+            }
+            dy_s[row] = dy;
           }
-          dy_s[row] = dy;
         }
       }
     }
   }
   __syncthreads();
 
-  // backward: dw_k = sum_i dY_i X[i,k] via MFMA (A row j=0 = dY), db = sum dY
-  const int ktiles = (K + 15) / 16;
-  for (int tk = 0; tk < ktiles; ++tk) {
-    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-    const int k = tk * 16 + r;
+  // backward: dw_k = sum_i dY_i X[i,k]; KT interleaved K-tiles; db = sum dY
+  {
+    f32x4 acc[KT];
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) acc[tk] = {0.f, 0.f, 0.f, 0.f};
     for (int i0 = 0; i0 < B; i0 += 4) {
       const int i = i0 + q;
       const float a = (r == 0 && i < B) ? dy_s[i] : 0.f;  // A[j=0][i]
-      const float b = (i < B && k < K) ? ldf(&X[(size_t)i * K + k]) : 0.f;
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) {
+        const int k = tk * 16 + r;
+        const float b = (i < B && k < K) ? xs[i * K + k] : 0.f;
+        acc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tk], 0, 0, 0);
+      }
     }
-    // D row j=0 lives in reg 0 of lanes with q==0; col = k
-    if (q == 0 && k < K) {
-      if (lr > 0.f) stf(&param[w_off + k], ldf(&param[w_off + k]) - lr * acc[0]);
-      else stf(&grad[w_off + k], acc[0]);
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      // D row j=0 lives in reg 0 of lanes with q==0; col = k
+      if (q == 0 && k < K) {
+        if (lr > 0.f) stf(&param[w_off + k], ws[k] - lr * acc[tk][0]);
+        else stf(&grad[w_off + k], acc[tk][0]);
+      }
     }
   }
   // db + loss reduce
@@ -696,6 +729,38 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
     if (lr > 0.f) stf(&param[b_off], bterm - lr * dbp);
     else stf(&grad[b_off], dbp);
     if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
+  }
+}
+
+template <typename T>
+static void launch_toy_fused(const torch::Tensor& x, const torch::Tensor& t,
+                             torch::Tensor& param_flat, torch::Tensor& grad_flat,
+                             float* lossp, bool use_mse, int w_off, int b_off,
+                             float lr, int B, int K) {
+  const T* xp = cdptr<T>(x);
+  const T* tp = cdptr<T>(t);
+  T* pp = dptr<T>(param_flat);
+  T* gp = dptr<T>(grad_flat);
+  auto go = [&](auto mt, auto kt) {
+    hipLaunchKernelGGL((k_toy_fused<T, decltype(mt)::value, decltype(kt)::value>),
+                       dim3(1), dim3(64), 0, cur_stream(), xp, tp, pp, gp,
+                       lossp, B, K, use_mse ? 1 : 0, w_off, b_off, lr);
+  };
+  using c1 = std::integral_constant<int, 1>;
+  using c2 = std::integral_constant<int, 2>;
+  using c4 = std::integral_constant<int, 4>;
+  using c8 = std::integral_constant<int, 8>;
+  const int mt = (B + 15) / 16, kt = (K + 15) / 16;
+  if (kt <= 1) {
+    if (mt <= 1) go(c1{}, c1{});
+    else if (mt <= 2) go(c2{}, c1{});
+    else if (mt <= 4) go(c4{}, c1{});
+    else go(c8{}, c1{});
+  } else {
+    if (mt <= 1) go(c1{}, c2{});
+    else if (mt <= 2) go(c2{}, c2{});
+    else if (mt <= 4) go(c4{}, c2{});
+    else go(c8{}, c2{});
   }
 }
 
@@ -712,11 +777,8 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
     lossp = loss_out.data_ptr<float>();
   }
   DISPATCH_F32_BF16(x.scalar_type(), "toy_fused_fwd_bwd", {
-    hipLaunchKernelGGL((k_toy_fused<scalar_t>), dim3(1), dim3(64), 0,
-                       cur_stream(), cdptr<scalar_t>(x), cdptr<scalar_t>(t),
-                       dptr<scalar_t>(param_flat), dptr<scalar_t>(grad_flat),
-                       lossp, B, K, use_mse ? 1 : 0,
-                       (int)w_off, (int)b_off, (float)lr);
+    launch_toy_fused<scalar_t>(x, t, param_flat, grad_flat, lossp, use_mse,
+                               (int)w_off, (int)b_off, (float)lr, B, K);
   });
   HIP_OK(hipGetLastError());
 }
