@@ -38,6 +38,9 @@ def parse_args():
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--dataset", type=int, default=2048)
     p.add_argument("--p50-probes", type=int, default=64)
+    p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
+                   help="compute dtype (fp32 = the reference's; bf16 = "
+                        "BASELINE.json config 2)")
     return p.parse_args()
 
 
@@ -46,10 +49,11 @@ class DeviceData:
     fast data path: one 164 KB upload, per-epoch permute+gather on device,
     per-step slicing is free)."""
 
-    def __init__(self, n, rank, world, batch, device, seed=1234):
+    def __init__(self, n, rank, world, batch, device, seed=1234,
+                 dtype=torch.float32):
         g = torch.Generator().manual_seed(seed)
-        self.X = torch.rand(n, 20, generator=g).to(device)
-        self.T = torch.rand(n, 1, generator=g).to(device)
+        self.X = torch.rand(n, 20, generator=g).to(device=device, dtype=dtype)
+        self.T = torch.rand(n, 1, generator=g).to(device=device, dtype=dtype)
         self.rank, self.world, self.batch = rank, world, batch
         self.per_rank = n // world
         self.steps_per_epoch = self.per_rank // batch
@@ -71,14 +75,14 @@ class DeviceData:
         return self._xs[lo:lo + self.batch], self._ts[lo:lo + self.batch]
 
 
-def build_engine(kind, comm, lr, device):
+def build_engine(kind, comm, lr, device, dtype=torch.float32):
     from mi355x_ddp.engine import GraphedToyStep, ToyFusedStep
     from mi355x_ddp.models import toy_model
     from mi355x_ddp.parallel import DDP, FusedSGD
     from mi355x_ddp import ops
 
     torch.manual_seed(4242)  # same init on every rank
-    model = toy_model(20, 1).to(device)
+    model = toy_model(20, 1).to(device=device, dtype=dtype)
 
     if kind == "autograd":
         engine = DDP(model, comm=comm)
@@ -125,9 +129,12 @@ def main():
         if use_cuda:
             torch.cuda.synchronize()
 
+    dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     step_fn = build_engine(args.engine if use_cuda else "autograd",
-                           comm, args.lr, device)
-    data = DeviceData(args.dataset, rank, world, args.batch, device)
+                           comm, args.lr, device,
+                           dtype if use_cuda else torch.float32)
+    data = DeviceData(args.dataset, rank, world, args.batch, device,
+                      dtype=dtype if use_cuda else torch.float32)
 
     # -- warmup (untimed) -------------------------------------------------
     for s in range(args.warmup):
@@ -181,7 +188,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,  # the reference publishes no numbers
-            "dtype": "fp32",
+            "dtype": args.dtype if use_cuda else "fp32",
             "data": "synthetic",
             "config": {
                 "model": "Linear(20,1)",
