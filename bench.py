@@ -32,8 +32,11 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=200)
-    p.add_argument("--engine", choices=["fused", "graph", "autograd"],
-                   default="fused")
+    p.add_argument("--engine",
+                   choices=["persistent", "fused", "graph", "autograd"],
+                   default="persistent",
+                   help="persistent = multi-step kernel at world 1 "
+                        "(falls back to fused when a comm exists)")
     p.add_argument("--batch", type=int, default=32, help="batch per rank")
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--dataset", type=int, default=2048)
@@ -76,7 +79,8 @@ class DeviceData:
 
 
 def build_engine(kind, comm, lr, device, dtype=torch.float32):
-    from mi355x_ddp.engine import GraphedToyStep, ToyFusedStep
+    from mi355x_ddp.engine import (GraphedToyStep, PersistentToyStep,
+                                   ToyFusedStep)
     from mi355x_ddp.models import toy_model
     from mi355x_ddp.parallel import DDP, FusedSGD
     from mi355x_ddp import ops
@@ -84,6 +88,7 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
     torch.manual_seed(4242)  # same init on every rank
     model = toy_model(20, 1).to(device=device, dtype=dtype)
 
+    noflush = lambda: None  # noqa: E731
     if kind == "autograd":
         engine = DDP(model, comm=comm)
         opt = FusedSGD(model.parameters(), lr=lr)
@@ -94,13 +99,16 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
             loss.backward()
             engine.finalize_backward()
             opt.step()
-        return step
+        return step, noflush
 
-    cls = GraphedToyStep if kind == "graph" else ToyFusedStep
+    if kind == "persistent" and comm is not None:
+        kind = "fused"  # multi-step kernel is the world-1 in-kernel-SGD path
+    cls = {"persistent": PersistentToyStep, "graph": GraphedToyStep,
+           "fused": ToyFusedStep}[kind]
     eng = cls(model, comm=comm, lr=lr, use_mse=True)
     if comm is not None:
         eng.reducer.broadcast_params(root=0)
-    return eng.step
+    return eng.step, getattr(eng, "flush", noflush)
 
 
 def main():
@@ -130,9 +138,9 @@ def main():
             torch.cuda.synchronize()
 
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    step_fn = build_engine(args.engine if use_cuda else "autograd",
-                           comm, args.lr, device,
-                           dtype if use_cuda else torch.float32)
+    step_fn, flush_fn = build_engine(args.engine if use_cuda else "autograd",
+                                     comm, args.lr, device,
+                                     dtype if use_cuda else torch.float32)
     data = DeviceData(args.dataset, rank, world, args.batch, device,
                       dtype=dtype if use_cuda else torch.float32)
 
@@ -140,13 +148,17 @@ def main():
     for s in range(args.warmup):
         x, t = data.batch_for(s)
         step_fn(x, t)
+    flush_fn()
     barrier()
 
-    # -- timed region: exactly K steps ------------------------------------
+    # -- timed region: exactly K steps (any deferred launches are flushed
+    #    INSIDE the bracket — all K steps' work executes before the
+    #    closing barrier+synchronize) ------------------------------------
     t0 = time.perf_counter()
     for s in range(args.steps):
         x, t = data.batch_for(args.warmup + s)
         step_fn(x, t)
+    flush_fn()
     barrier()
     elapsed = time.perf_counter() - t0
 
@@ -166,6 +178,7 @@ def main():
             torch.cuda.synchronize()
         t1 = time.perf_counter()
         step_fn(x, t)
+        flush_fn()  # deferred engines: probe = true single-step latency
         if use_cuda:
             torch.cuda.synchronize()
         probes.append(time.perf_counter() - t1)
@@ -196,7 +209,9 @@ def main():
                 "in_features": 20,
                 "dataset_size": args.dataset,
                 "parallelism": f"dp{world}",
-                "engine": args.engine if use_cuda else "autograd-cpu",
+                "engine": ("autograd-cpu" if not use_cuda else
+                           "fused" if args.engine == "persistent" and world > 1
+                           else args.engine),
                 "loss": "mse",
                 "p50_step_ms": p50_ms,
             },

@@ -69,6 +69,71 @@ class ToyFusedStep:
         ops.ext().sgd_flat(self.flat_param, self.flat_grad, self.lr, True)
 
 
+class PersistentToyStep(ToyFusedStep):
+    """World-1 engine: runs of consecutive steps execute as ONE multi-step
+    kernel launch (`toy_multistep`), weights resident in LDS across steps.
+
+    `step(x, t)` defers when the incoming batch is the next contiguous
+    [B, K] slice of the same device buffer (the device-resident epoch
+    shard, bench.py DeviceData); any other batch flushes the pending run
+    and starts a new one. Per-step arithmetic is bitwise-identical to the
+    single-step fused kernel, so deferral changes WHEN work is launched,
+    never what is computed. Callers must call `flush()` before
+    synchronizing the stream for timing/reading params — bench.py does
+    before every barrier. Deferred-run length is capped by `max_defer`.
+
+    With track_loss=True, loss_out holds the LAST executed step's loss.
+    """
+
+    def __init__(self, *args, max_defer: int = 1024, **kwargs):
+        super().__init__(*args, **kwargs)
+        assert self.comm is None, \
+            "PersistentToyStep is the world-1 path (in-kernel SGD)"
+        self.max_defer = max_defer
+        self._x0 = None        # first batch of the pending run (keeps storage)
+        self._t0 = None
+        self._count = 0
+
+    def _next_ptrs(self):
+        bk = self._x0.shape[0] * self._x0.shape[1] * self._x0.element_size()
+        bt = self._t0.numel() * self._t0.element_size()
+        return (self._x0.data_ptr() + self._count * bk,
+                self._t0.data_ptr() + self._count * bt)
+
+    def step(self, x: torch.Tensor, t: torch.Tensor) -> None:
+        if self._count > 0:
+            nx, nt = self._next_ptrs()
+            if (x.data_ptr() == nx and t.data_ptr() == nt
+                    and x.shape == self._x0.shape and t.shape == self._t0.shape
+                    and x.is_contiguous() and t.is_contiguous()):
+                self._count += 1
+                if self._count >= self.max_defer:
+                    self.flush()
+                return
+            self.flush()
+        if not (x.is_cuda and x.is_contiguous() and t.is_contiguous()):
+            super().step(x, t)
+            return
+        self._x0, self._t0, self._count = x, t, 1
+
+    def flush(self) -> None:
+        if self._count == 0:
+            return
+        x0, t0, n = self._x0, self._t0, self._count
+        self._x0 = self._t0 = None
+        self._count = 0
+        B, K = x0.shape
+        if n == 1:
+            super().step(x0, t0)
+            return
+        xall = x0.as_strided((n * B, K), (K, 1))
+        tall = t0.as_strided((n * B,) + t0.shape[1:],
+                             (t0.stride(0),) + t0.stride()[1:])
+        ops.ext().toy_multistep(xall, tall, self.flat_param, self.loss_out,
+                                self.use_mse, self.w_off, self.b_off,
+                                self.lr, B)
+
+
 class GraphedToyStep(ToyFusedStep):
     """ToyFusedStep captured in a hipGraph.
 

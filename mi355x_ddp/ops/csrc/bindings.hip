@@ -36,6 +36,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("t"), py::arg("param_flat"), py::arg("grad_flat"),
         py::arg("loss_out"), py::arg("use_mse") = true,
         py::arg("w_off") = 0, py::arg("b_off") = 0, py::arg("lr") = 0.0);
+  m.def("toy_multistep", &mi355x::toy_multistep, py::arg("x"), py::arg("t"),
+        py::arg("param_flat"), py::arg("loss_out"), py::arg("use_mse") = true,
+        py::arg("w_off") = 0, py::arg("b_off") = 0, py::arg("lr") = 0.0,
+        py::arg("batch") = 32);
 
   py::class_<mi355x::RcclComm>(m, "RcclComm")
       .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),

@@ -732,6 +732,229 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
   }
 }
 
+// ---------------------------------------------------------------------------
+// Multi-step persistent toy trainer: S sequential SGD steps in ONE launch.
+//
+// The single-step kernel above executes in ~6 us wall: ~0.3 us of MFMA and
+// ~5+ us of dispatch ramp + HBM round trips for 84 B of params re-read and
+// re-written every launch. Since the epoch shard is device-resident and
+// consecutive steps read consecutive [B,K] slices, this kernel keeps the
+// weights in LDS across steps, double-buffers the next step's X/T tile
+// through registers while MFMAing the current one, and only touches HBM
+// for the streamed batches plus one final param write-back. Per-step
+// arithmetic (MFMA tiling, update order, bf16 rounding of the stored
+// params) is IDENTICAL to k_toy_fused, so S multi-steps == S single-step
+// launches bitwise; tests/test_engine_gpu.py holds it to that.
+//
+// World-1 only (lr applied in-kernel; no collective exists). B<=128, K<=32.
+// ---------------------------------------------------------------------------
+template <typename T>
+__device__ __forceinline__ float round_store(float v);
+template <>
+__device__ __forceinline__ float round_store<float>(float v) { return v; }
+template <>
+__device__ __forceinline__ float round_store<__hip_bfloat16>(float v) {
+  return __bfloat162float(__float2bfloat16(v));
+}
+
+template <typename T, int MT, int KT>
+__global__ void k_toy_multistep(const T* __restrict__ X,
+                                const T* __restrict__ Tg,
+                                T* __restrict__ param,
+                                float* __restrict__ loss_out,
+                                int B, int K, int S, int use_mse,
+                                int w_off, int b_off, float lr) {
+  constexpr int NR = (MT * 16 * KT * 16 + 63) / 64;  // X regs per lane
+  constexpr int NT = (MT * 16 + 63) / 64;            // target regs per lane
+  const int lane = threadIdx.x;
+  const int r = lane & 15, q = lane >> 4;
+  __shared__ float xs2[2][128 * 32];
+  __shared__ float ts2[2][128];
+  __shared__ float ws[33];  // w (K) + bias at ws[32]; persists across steps
+  __shared__ float dy_s[128];
+
+  if (lane < K) ws[lane] = ldf(&param[w_off + lane]);
+  if (lane == K) ws[32] = ldf(&param[b_off]);
+
+  const int total = B * K;
+  float xr[NR], tr[NT];
+  // prologue: step 0 tile -> regs
+#pragma unroll
+  for (int j = 0; j < NR; ++j) {
+    const int i = lane + j * 64;
+    if (i < total) xr[j] = ldf(&X[i]);
+  }
+#pragma unroll
+  for (int j = 0; j < NT; ++j) {
+    const int i = lane + j * 64;
+    if (i < B) tr[j] = ldf(&Tg[i]);
+  }
+
+  float loss_last = 0.f;
+  for (int s = 0; s < S; ++s) {
+    const int buf = s & 1;
+    float* xs = xs2[buf];
+    float* ts = ts2[buf];
+    // stage this step's regs into LDS
+#pragma unroll
+    for (int j = 0; j < NR; ++j) {
+      const int i = lane + j * 64;
+      if (i < total) xs[i] = xr[j];
+    }
+#pragma unroll
+    for (int j = 0; j < NT; ++j) {
+      const int i = lane + j * 64;
+      if (i < B) ts[i] = tr[j];
+    }
+    __syncthreads();
+    // issue next step's global loads: latency hides under this step's MFMAs
+    if (s + 1 < S) {
+      const T* Xn = X + (size_t)(s + 1) * total;
+      const T* Tn = Tg + (size_t)(s + 1) * B;
+#pragma unroll
+      for (int j = 0; j < NR; ++j) {
+        const int i = lane + j * 64;
+        if (i < total) xr[j] = ldf(&Xn[i]);
+      }
+#pragma unroll
+      for (int j = 0; j < NT; ++j) {
+        const int i = lane + j * 64;
+        if (i < B) tr[j] = ldf(&Tn[i]);
+      }
+    }
+
+    // ---- forward (identical tiling to k_toy_fused) ----
+    const float bterm = ws[32];
+    float loss_acc = 0.f;
+    {
+      f32x4 acc[MT];
+#pragma unroll
+      for (int tm = 0; tm < MT; ++tm) acc[tm] = {0.f, 0.f, 0.f, 0.f};
+      for (int k0 = 0; k0 < K; k0 += 4) {
+        const int k = k0 + q;
+        const float b = (r == 0 && k < K) ? ws[k] : 0.f;
+#pragma unroll
+        for (int tm = 0; tm < MT; ++tm) {
+          const int m = tm * 16 + r;
+          const float a = (m < B && k < K) ? xs[m * K + k] : 0.f;
+          acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tm], 0, 0, 0);
+        }
+      }
+#pragma unroll
+      for (int tm = 0; tm < MT; ++tm) {
+        if (r == 0) {
+#pragma unroll
+          for (int i = 0; i < 4; ++i) {
+            const int row = tm * 16 + q * 4 + i;
+            if (row < B) {
+              const float y = acc[tm][i] + bterm;
+              float dy;
+              if (use_mse) {
+                const float d = y - ts[row];
+                loss_acc += d * d;
+                dy = 2.f * d / (float)B;
+              } else {
+                dy = 0.f;  // degenerate 1-logit CE
+              }
+              dy_s[row] = dy;
+            }
+          }
+        }
+      }
+    }
+    __syncthreads();  // dy_s visible to all lanes before the bwd MFMAs
+
+    // ---- backward + in-LDS SGD update ----
+    {
+      f32x4 acc[KT];
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) acc[tk] = {0.f, 0.f, 0.f, 0.f};
+      for (int i0 = 0; i0 < B; i0 += 4) {
+        const int i = i0 + q;
+        const float a = (r == 0 && i < B) ? dy_s[i] : 0.f;
+#pragma unroll
+        for (int tk = 0; tk < KT; ++tk) {
+          const int k = tk * 16 + r;
+          const float b = (i < B && k < K) ? xs[i * K + k] : 0.f;
+          acc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tk], 0, 0, 0);
+        }
+      }
+      __syncthreads();  // all ws reads of this step done before the update
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) {
+        const int k = tk * 16 + r;
+        if (q == 0 && k < K)
+          ws[k] = round_store<T>(ws[k] - lr * acc[tk][0]);
+      }
+    }
+    float dbp = 0.f;
+    for (int i = lane; i < B; i += 64) dbp += dy_s[i];
+    dbp = wave_sum(dbp);
+    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
+    if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
+    __syncthreads();
+  }
+
+  // final param write-back (one HBM trip for the whole S-step run)
+  if (lane < K) stf(&param[w_off + lane], ws[lane]);
+  if (lane == K) stf(&param[b_off], ws[32]);
+  if (lane == 0 && loss_out) *loss_out = use_mse ? loss_last / (float)B : 0.f;
+}
+
+template <typename T>
+static void launch_toy_multistep(const torch::Tensor& x, const torch::Tensor& t,
+                                 torch::Tensor& param_flat, float* lossp,
+                                 bool use_mse, int w_off, int b_off, float lr,
+                                 int B, int K, int S) {
+  const T* xp = cdptr<T>(x);
+  const T* tp = cdptr<T>(t);
+  T* pp = dptr<T>(param_flat);
+  auto go = [&](auto mt, auto kt) {
+    hipLaunchKernelGGL((k_toy_multistep<T, decltype(mt)::value, decltype(kt)::value>),
+                       dim3(1), dim3(64), 0, cur_stream(), xp, tp, pp, lossp,
+                       B, K, S, use_mse ? 1 : 0, w_off, b_off, lr);
+  };
+  using c1 = std::integral_constant<int, 1>;
+  using c2 = std::integral_constant<int, 2>;
+  using c4 = std::integral_constant<int, 4>;
+  using c8 = std::integral_constant<int, 8>;
+  const int mt = (B + 15) / 16, kt = (K + 15) / 16;
+  if (kt <= 1) {
+    if (mt <= 1) go(c1{}, c1{});
+    else if (mt <= 2) go(c2{}, c1{});
+    else if (mt <= 4) go(c4{}, c1{});
+    else go(c8{}, c1{});
+  } else {
+    if (mt <= 1) go(c1{}, c2{});
+    else if (mt <= 2) go(c2{}, c2{});
+    else if (mt <= 4) go(c4{}, c2{});
+    else go(c8{}, c2{});
+  }
+}
+
+void toy_multistep(torch::Tensor x, torch::Tensor t, torch::Tensor param_flat,
+                   torch::Tensor loss_out, bool use_mse,
+                   int64_t w_off, int64_t b_off, double lr, int64_t batch) {
+  const int B = (int)batch, K = (int)x.size(1);
+  TORCH_CHECK(B <= 128 && K <= 32, "toy multistep supports B<=128, K<=32");
+  TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
+  TORCH_CHECK(x.size(0) % B == 0, "x rows must be a multiple of batch");
+  TORCH_CHECK(t.size(0) == x.size(0), "t rows must match x rows");
+  TORCH_CHECK(lr > 0.0, "toy_multistep is the world-1 in-kernel-SGD path");
+  const int S = (int)(x.size(0) / B);
+  if (S == 0) return;
+  float* lossp = nullptr;
+  if (loss_out.defined() && loss_out.numel()) {
+    TORCH_CHECK(loss_out.scalar_type() == at::kFloat, "loss_out must be f32");
+    lossp = loss_out.data_ptr<float>();
+  }
+  DISPATCH_F32_BF16(x.scalar_type(), "toy_multistep", {
+    launch_toy_multistep<scalar_t>(x, t, param_flat, lossp, use_mse,
+                                   (int)w_off, (int)b_off, (float)lr, B, K, S);
+  });
+  HIP_OK(hipGetLastError());
+}
+
 template <typename T>
 static void launch_toy_fused(const torch::Tensor& x, const torch::Tensor& t,
                              torch::Tensor& param_flat, torch::Tensor& grad_flat,

@@ -62,4 +62,13 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor loss_out, bool use_mse,
                        int64_t w_off, int64_t b_off, double lr);
 
+// Multi-step persistent toy trainer (world-1): x is [S*batch, K] of S
+// consecutive batches; runs S full fwd+loss+bwd+SGD steps in ONE launch
+// with the weights resident in LDS (bitwise-identical per-step arithmetic
+// to toy_fused_fwd_bwd with in-kernel SGD). loss_out, when non-empty,
+// receives the LAST step's loss.
+void toy_multistep(torch::Tensor x, torch::Tensor t, torch::Tensor param_flat,
+                   torch::Tensor loss_out, bool use_mse,
+                   int64_t w_off, int64_t b_off, double lr, int64_t batch);
+
 }  // namespace mi355x

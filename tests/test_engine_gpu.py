@@ -72,6 +72,72 @@ def test_graphed_matches_fused():
     assert torch.allclose(b, b_ref, atol=1e-5)
 
 
+def _train_persistent(dtype=torch.float32, scattered=False):
+    from mi355x_ddp.engine import PersistentToyStep
+    from mi355x_ddp.models import toy_model
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(device=DEV, dtype=dtype)
+    eng = PersistentToyStep(model, comm=None, lr=LR, use_mse=True)
+    X, T = _data()
+    X, T = X.to(dtype), T.to(dtype)
+    if scattered:
+        # non-contiguous feeding order of fresh tensors: every step falls
+        # back to the eager single-step kernel
+        for s in range(STEPS):
+            eng.step(X[s].clone(), T[s].clone())
+    else:
+        # contiguous shard feeding, as DeviceData produces: one flush runs
+        # the whole run through the multi-step kernel
+        Xf = X.reshape(STEPS * 32, 20).contiguous()
+        Tf = T.reshape(STEPS * 32, 1).contiguous()
+        for s in range(STEPS):
+            eng.step(Xf[s * 32:(s + 1) * 32], Tf[s * 32:(s + 1) * 32])
+    eng.flush()
+    torch.cuda.synchronize()
+    return (model.weight.detach().float().cpu().clone(),
+            model.bias.detach().float().cpu().clone())
+
+
+def _train_fused_dtype(dtype):
+    from mi355x_ddp.engine import ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(device=DEV, dtype=dtype)
+    eng = ToyFusedStep(model, comm=None, lr=LR, use_mse=True)
+    X, T = _data()
+    X, T = X.to(dtype), T.to(dtype)
+    for s in range(STEPS):
+        eng.step(X[s].contiguous(), T[s].contiguous())
+    torch.cuda.synchronize()
+    return (model.weight.detach().float().cpu().clone(),
+            model.bias.detach().float().cpu().clone())
+
+
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_persistent_multistep_bitwise_matches_single_step(dtype):
+    # S deferred steps in ONE kernel == S single-step launches, bitwise:
+    # the multi-step kernel uses identical per-step arithmetic (including
+    # bf16 round-trip of the stored params between steps).
+    w_ref, b_ref = _train_fused_dtype(dtype)
+    w, b = _train_persistent(dtype=dtype)
+    assert torch.equal(w, w_ref), (w - w_ref).abs().max()
+    assert torch.equal(b, b_ref)
+
+
+def test_persistent_fallback_scattered_batches():
+    w_ref, b_ref = _train_fused_dtype(torch.float32)
+    w, b = _train_persistent(scattered=True)
+    assert torch.equal(w, w_ref)
+    assert torch.equal(b, b_ref)
+
+
+def test_persistent_matches_autograd():
+    w_ref, b_ref = _train_autograd()
+    w, b = _train_persistent()
+    assert torch.allclose(w, w_ref, atol=1e-5), (w - w_ref).abs().max()
+    assert torch.allclose(b, b_ref, atol=1e-5)
+
+
 def test_bench_single_gpu_json():
     root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
     out = subprocess.run(
