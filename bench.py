@@ -68,14 +68,22 @@ class DeviceData:
     def batch_for(self, step):
         epoch, s = divmod(step, self.steps_per_epoch)
         if epoch != self._epoch:
-            g = torch.Generator(device="cpu").manual_seed(1000 + epoch)
-            perm = torch.randperm(self.X.shape[0], generator=g).to(self.device)
+            # epoch-seeded permutation generated ON DEVICE (identical on
+            # every rank for a given epoch; no host round trip)
+            g = torch.Generator(device=self.device).manual_seed(1000 + epoch)
+            perm = torch.randperm(self.X.shape[0], generator=g,
+                                  device=self.device)
             shard = perm[self.rank::self.world][: self.per_rank]
             self._xs = self.X[shard].contiguous()
             self._ts = self.T[shard].contiguous()
+            b = self.batch
+            # pre-slice the epoch's batch views once: per-step cost is a
+            # list index, not two tensor-slice constructions
+            self._views = [(self._xs[i * b:(i + 1) * b],
+                            self._ts[i * b:(i + 1) * b])
+                           for i in range(self.steps_per_epoch)]
             self._epoch = epoch
-        lo = s * self.batch
-        return self._xs[lo:lo + self.batch], self._ts[lo:lo + self.batch]
+        return self._views[s]
 
 
 def build_engine(kind, comm, lr, device, dtype=torch.float32):

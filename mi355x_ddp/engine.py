@@ -93,20 +93,20 @@ class PersistentToyStep(ToyFusedStep):
         self._x0 = None        # first batch of the pending run (keeps storage)
         self._t0 = None
         self._count = 0
-
-    def _next_ptrs(self):
-        bk = self._x0.shape[0] * self._x0.shape[1] * self._x0.element_size()
-        bt = self._t0.numel() * self._t0.element_size()
-        return (self._x0.data_ptr() + self._count * bk,
-                self._t0.data_ptr() + self._count * bt)
+        self._nx = 0           # expected data_ptr of the next contiguous batch
+        self._nt = 0
+        self._bk = 0           # batch strides in bytes
+        self._bt = 0
+        self._shape = None
 
     def step(self, x: torch.Tensor, t: torch.Tensor) -> None:
         if self._count > 0:
-            nx, nt = self._next_ptrs()
-            if (x.data_ptr() == nx and t.data_ptr() == nt
-                    and x.shape == self._x0.shape and t.shape == self._t0.shape
-                    and x.is_contiguous() and t.is_contiguous()):
+            # fast path: the next contiguous slice of the pending run
+            if (x.data_ptr() == self._nx and t.data_ptr() == self._nt
+                    and x.shape == self._shape):
                 self._count += 1
+                self._nx += self._bk
+                self._nt += self._bt
                 if self._count >= self.max_defer:
                     self.flush()
                 return
@@ -115,6 +115,11 @@ class PersistentToyStep(ToyFusedStep):
             super().step(x, t)
             return
         self._x0, self._t0, self._count = x, t, 1
+        self._shape = x.shape
+        self._bk = x.numel() * x.element_size()
+        self._bt = t.numel() * t.element_size()
+        self._nx = x.data_ptr() + self._bk
+        self._nt = t.data_ptr() + self._bt
 
     def flush(self) -> None:
         if self._count == 0:

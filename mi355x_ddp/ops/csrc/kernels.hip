@@ -654,6 +654,7 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
 
   // forward: y = X @ w + b; MT interleaved 16-row tiles, j=0 column only
   const float bterm = ws[32];
+  const float inv2B = 2.f / (float)B;
   float loss_acc = 0.f;
   {
     f32x4 acc[MT];
@@ -681,7 +682,7 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
             if (use_mse) {
               const float d = y - ts[row];
               loss_acc += d * d;
-              dy = 2.f * d / (float)B;
+              dy = d * inv2B;
             } else {
               // CE over one logit: log_softmax == 0 -> loss == 0, dY == 0
               // (the reference's degenerate loss).
@@ -825,6 +826,7 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
 
     // ---- forward (identical tiling to k_toy_fused) ----
     const float bterm = ws[32];
+    const float inv2B = 2.f / (float)B;
     float loss_acc = 0.f;
     {
       f32x4 acc[MT];
@@ -852,7 +854,7 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
               if (use_mse) {
                 const float d = y - ts[row];
                 loss_acc += d * d;
-                dy = 2.f * d / (float)B;
+                dy = d * inv2B;
               } else {
                 dy = 0.f;  // degenerate 1-logit CE
               }
@@ -901,6 +903,173 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
   if (lane == 0 && loss_out) *loss_out = use_mse ? loss_last / (float)B : 0.f;
 }
 
+// ---------------------------------------------------------------------------
+// Compile-time-shape multi-step trainer for the canonical toy shape.
+// All loop bounds are template constants so every MFMA operand lives in a
+// REGISTER, prefetched from global memory in operand layout during the
+// previous step's compute (the epoch shard is L2-resident after its
+// gather, so the depth-1 prefetch covers the latency). LDS is used only
+// where data must cross lanes: the w vector (updated by q==0 lanes, read
+// by r==0 lanes) and the per-row dY exchange between the forward readout
+// and the backward MFMA. Divisions are hoisted to one reciprocal
+// (2/B is exact for the reference batch 32, so dy is bitwise-unchanged).
+// ---------------------------------------------------------------------------
+template <typename T, int B_, int K_>
+__global__ void __launch_bounds__(64, 1)
+k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
+                     T* __restrict__ param, float* __restrict__ loss_out,
+                     int S, int use_mse, int w_off, int b_off, float lr) {
+  constexpr int MT = (B_ + 15) / 16;   // fwd 16-row tiles
+  constexpr int KT = (K_ + 15) / 16;   // bwd 16-col tiles
+  constexpr int KS = (K_ + 3) / 4;     // fwd k-steps
+  constexpr int BS = (B_ + 3) / 4;     // bwd i-steps
+  const int lane = threadIdx.x;
+  const int r = lane & 15, q = lane >> 4;
+  __shared__ float ws[33];             // w (K_) + bias at ws[32]
+  __shared__ float dy_s[MT * 16];
+
+  if (lane < K_) ws[lane] = ldf(&param[w_off + lane]);
+  if (lane == K_) ws[32] = ldf(&param[b_off]);
+  const float inv2B = 2.f / (float)B_;
+
+  // Per-lane operand registers, software-pipelined: `c*` hold the step
+  // being computed, `n*` receive the next step's loads (issued at the top
+  // of the iteration, consumed — behind one vmcnt wait — at its end).
+  // All loads are UNCONDITIONAL with clamped addresses + value selects:
+  // ternary-guarded loads compile to exec-masked branch-per-element code
+  // that serializes the burst (seen in the r01b ISA), selects do not.
+  float cfA[MT][KS];  // fwd A: X[tm*16+r][4*kk+q]
+  float cbB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
+  float ctR[MT][4];   // targets for rows tm*16+q*4+i
+
+  auto prefetch = [&](int s) {
+    const T* Xs = X + (size_t)s * (B_ * K_);
+    const T* Ts = Tg + (size_t)s * B_;
+    // Loads are raw (address-clamped, no value select): out-of-range
+    // lanes load in-bounds garbage whose products are either multiplied
+    // by a zeroed LDS-side operand (wv/av) or discarded by the guarded
+    // ws/dy writes — so no per-load mask, and therefore no vmcnt wait
+    // until the next iteration's first MFMA use.
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) {
+      const int m = tm * 16 + r;
+      const int mc = (m < B_) ? m : B_ - 1;
+#pragma unroll
+      for (int kk = 0; kk < KS; ++kk) {
+        const int k = 4 * kk + q;
+        const int kc = (k < K_) ? k : K_ - 1;
+        cfA[tm][kk] = ldf(&Xs[mc * K_ + kc]);
+      }
+    }
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      const int kc = (k < K_) ? k : K_ - 1;
+#pragma unroll
+      for (int ii = 0; ii < BS; ++ii) {
+        const int i = 4 * ii + q;
+        const int ic = (i < B_) ? i : B_ - 1;
+        cbB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
+      }
+    }
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = tm * 16 + q * 4 + i;
+        ctR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
+      }
+  };
+
+  prefetch(0);
+  float loss_last = 0.f;
+  for (int s = 0; s < S; ++s) {
+    // batch the w reads (LDS; one unconditional read + select per kk)
+    float wv[KS];
+#pragma unroll
+    for (int kk = 0; kk < KS; ++kk) {
+      const int k = 4 * kk + q;
+      const float v = ws[(k < K_) ? k : K_ - 1];
+      wv[kk] = (r == 0 && k < K_) ? v : 0.f;
+    }
+    const float bterm = ws[32];
+
+    // ---- forward: y = X @ w + b ----
+    f32x4 acc[MT];
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) acc[tm] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kk = 0; kk < KS; ++kk)
+#pragma unroll
+      for (int tm = 0; tm < MT; ++tm)
+        acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(cfA[tm][kk], wv[kk],
+                                                       acc[tm], 0, 0, 0);
+
+    // ---- loss grad (rows live on r==0 lanes) + dY exchange ----
+    float loss_acc = 0.f;
+    if (r == 0) {
+#pragma unroll
+      for (int tm = 0; tm < MT; ++tm)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int row = tm * 16 + q * 4 + i;
+          float dy = 0.f;
+          if (row < B_ && use_mse) {
+            const float d = acc[tm][i] + bterm - ctR[tm][i];
+            loss_acc += d * d;
+            dy = d * inv2B;
+          }
+          if (row < B_) dy_s[row] = dy;
+        }
+    }
+    __syncthreads();  // dy_s visible to all lanes
+
+    // ---- backward: dw_k = sum_i dY_i X[i,k] ----
+    float av[BS];
+#pragma unroll
+    for (int ii = 0; ii < BS; ++ii) {
+      const int i = 4 * ii + q;
+      const float v = dy_s[(i < B_) ? i : B_ - 1];
+      av[ii] = (r == 0 && i < B_) ? v : 0.f;
+    }
+    f32x4 gacc[KT];
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) gacc[tk] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int ii = 0; ii < BS; ++ii)
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk)
+        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], cbB[tk][ii],
+                                                        gacc[tk], 0, 0, 0);
+
+    // cur regs are dead from here: load next step's operands straight into
+    // them — the vmcnt wait attaches to their first use (next iteration's
+    // forward MFMA), shadowed by the update/barrier below
+    if (s + 1 < S) prefetch(s + 1);
+
+    // db: same lane-strided dy_s reduction order as the single-step
+    // kernel (keeps the multi-step history bitwise-identical)
+    float dbp = 0.f;
+#pragma unroll
+    for (int i = lane; i < B_; i += 64) dbp += dy_s[i];
+    dbp = wave_sum(dbp);
+    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
+
+    __syncthreads();  // all ws/dy reads done before the update
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      if (q == 0 && k < K_) ws[k] = round_store<T>(ws[k] - lr * gacc[tk][0]);
+    }
+    if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
+    __syncthreads();
+  }
+
+  if (lane < K_) stf(&param[w_off + lane], ws[lane]);
+  if (lane == K_) stf(&param[b_off], ws[32]);
+  if (lane == 0 && loss_out) *loss_out = use_mse ? loss_last / (float)B_ : 0.f;
+}
+
 template <typename T>
 static void launch_toy_multistep(const torch::Tensor& x, const torch::Tensor& t,
                                  torch::Tensor& param_flat, float* lossp,
@@ -909,6 +1078,12 @@ static void launch_toy_multistep(const torch::Tensor& x, const torch::Tensor& t,
   const T* xp = cdptr<T>(x);
   const T* tp = cdptr<T>(t);
   T* pp = dptr<T>(param_flat);
+  if (B == 32 && K == 20) {  // the reference shape: compile-time fast path
+    hipLaunchKernelGGL((k_toy_multistep_spec<T, 32, 20>), dim3(1), dim3(64),
+                       0, cur_stream(), xp, tp, pp, lossp, S,
+                       use_mse ? 1 : 0, w_off, b_off, lr);
+    return;
+  }
   auto go = [&](auto mt, auto kt) {
     hipLaunchKernelGGL((k_toy_multistep<T, decltype(mt)::value, decltype(kt)::value>),
                        dim3(1), dim3(64), 0, cur_stream(), xp, tp, pp, lossp,
