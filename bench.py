@@ -61,26 +61,51 @@ class DeviceData:
         self.per_rank = n // world
         self.steps_per_epoch = self.per_rank // batch
         self.device = device
+        self._cuda = device.type == "cuda"
         self._epoch = -1
-        self._xs = None
-        self._ts = None
+        self._gen = (torch.Generator(device=device) if self._cuda
+                     else torch.Generator())
+        # epoch shards are gathered on a SIDE stream one epoch ahead, so the
+        # sort+gather overlaps the previous epoch's training kernels
+        self._side = torch.cuda.Stream() if self._cuda else None
+        self._pending = {}  # epoch -> (xs, ts, ready_event)
+
+    def _gather(self, epoch):
+        # epoch-seeded permutation generated ON DEVICE (identical on every
+        # rank for a given epoch; no host round trip)
+        self._gen.manual_seed(1000 + epoch)
+        perm = torch.randperm(self.X.shape[0], generator=self._gen,
+                              device=self.device)
+        shard = perm[self.rank::self.world][: self.per_rank]
+        return self.X[shard].contiguous(), self.T[shard].contiguous()
+
+    def _prep(self, epoch):
+        if not self._cuda:
+            self._pending[epoch] = (*self._gather(epoch), None)
+            return
+        with torch.cuda.stream(self._side):
+            xs, ts = self._gather(epoch)
+            ev = torch.cuda.Event()
+            ev.record(self._side)
+        self._pending[epoch] = (xs, ts, ev)
 
     def batch_for(self, step):
         epoch, s = divmod(step, self.steps_per_epoch)
         if epoch != self._epoch:
-            # epoch-seeded permutation generated ON DEVICE (identical on
-            # every rank for a given epoch; no host round trip)
-            g = torch.Generator(device=self.device).manual_seed(1000 + epoch)
-            perm = torch.randperm(self.X.shape[0], generator=g,
-                                  device=self.device)
-            shard = perm[self.rank::self.world][: self.per_rank]
-            self._xs = self.X[shard].contiguous()
-            self._ts = self.T[shard].contiguous()
+            if epoch not in self._pending:
+                self._prep(epoch)
+            xs, ts, ev = self._pending.pop(epoch)
+            if ev is not None:
+                main = torch.cuda.current_stream()
+                main.wait_event(ev)
+                # shards are allocated on the side stream but consumed on the
+                # main stream: mark the usage so the caching allocator only
+                # reuses their blocks after main-stream kernels finish
+                xs.record_stream(main)
+                ts.record_stream(main)
+            self._prep(epoch + 1)  # prefetch next epoch on the side stream
             b = self.batch
-            # pre-slice the epoch's batch views once: per-step cost is a
-            # list index, not two tensor-slice constructions
-            self._views = [(self._xs[i * b:(i + 1) * b],
-                            self._ts[i * b:(i + 1) * b])
+            self._views = [(xs[i * b:(i + 1) * b], ts[i * b:(i + 1) * b])
                            for i in range(self.steps_per_epoch)]
             self._epoch = epoch
         return self._views[s]

@@ -43,32 +43,43 @@ def test_resnet50_train_steps_multibucket():
 def test_resnet50_grads_match_plain_torch():
     # one backward through the engine == plain autograd grads. Both runs use
     # the framework CE kernel so the comparison isolates the reducer's
-    # rebinding (flat param/grad views); tolerance is norm-relative because
-    # MIOpen conv-backward uses atomics and is not bitwise repeatable.
+    # rebinding (flat param/grad views). MIOpen conv-backward picks
+    # algorithms per pointer alignment and uses atomics, so the bound is
+    # the measured run-to-run noise floor of the PLAIN path, with margin.
     torch.manual_seed(1)
     model = resnet50().to(DEV)
     x = torch.rand(4, 3, 224, 224, device=DEV)
     t = torch.softmax(torch.rand(4, 1000, device=DEV), dim=1)
-    bn_state = {k: v.clone() for k, v in model.state_dict().items()}
+    state0 = {k: v.clone() for k, v in model.state_dict().items()}
 
-    # plain reference first (before reducer rebinding)
-    y = model(x)
-    loss = ops.cross_entropy(y, t)
-    loss.backward()
-    ref_grads = [p.grad.detach().clone() for p in model.parameters()]
-    for p in model.parameters():
-        p.grad = None
-    model.load_state_dict(bn_state)  # rewind BN running stats
+    def run_plain():
+        y = model(x)
+        loss = ops.cross_entropy(y, t)
+        loss.backward()
+        gs = [p.grad.detach().clone() for p in model.parameters()]
+        for p in model.parameters():
+            p.grad = None
+        model.load_state_dict(state0)  # rewind BN running stats
+        return loss.detach(), gs
+
+    loss_a, ref = run_plain()
+    _, ref2 = run_plain()  # noise floor of the plain path itself
 
     reducer = Reducer(list(model.parameters()), comm=None, bucket_cap_mb=25.0)
     y2 = model(x)
     loss2 = ops.cross_entropy(y2, t)
     loss2.backward()
     reducer.finalize()
-    assert torch.allclose(loss2, loss, atol=1e-4, rtol=1e-4)
-    for p, ref in zip(model.parameters(), ref_grads):
-        assert p.grad is not None
-        num = (p.grad - ref).norm()
-        den = ref.norm() + 1e-8
-        assert num / den < 1e-2, (p.shape, float(num / den),
-                                  (p.grad - ref).abs().max())
+    assert torch.allclose(loss2, loss_a, atol=1e-4, rtol=1e-4)
+
+    def rel(a, b):
+        return float((a - b).norm() / (b.norm() + 1e-12))
+
+    noise = max(rel(g2, g1) for g1, g2 in zip(ref, ref2))
+    bad = []
+    for (name, p), g1 in zip(model.named_parameters(), ref):
+        r = rel(p.grad, g1)
+        if r > max(20 * noise, 2e-3):
+            bad.append((name, tuple(p.shape), r))
+    assert not bad, (f"noise_floor={noise:.2e}; "
+                     f"{len(bad)} params exceed bound: {bad[:8]}")
