@@ -100,6 +100,10 @@ class Trainer:
         self.epochs_run = 0
         self.profile = profile
         self.profile_dir = profile_dir
+        # copy-stream H2D prefetch (on by default on GPU; MI355X_PREFETCH=0
+        # falls back to inline non_blocking copies)
+        self._use_prefetcher = (self.device.type == "cuda"
+                                and os.environ.get("MI355X_PREFETCH", "1") != "0")
 
         self.model = model.to(self.device)
 
@@ -192,10 +196,19 @@ class Trainer:
         sampler = getattr(self.train_data, "sampler", None)
         if sampler is not None and hasattr(sampler, "set_epoch"):
             sampler.set_epoch(epoch)
+        if self._use_prefetcher:
+            # copy-stream pipeline: batches arrive already on device,
+            # uploaded one step ahead (fixes the reference's
+            # synchronous .to(), multigpu.py:49-50)
+            from .data import DevicePrefetcher
+            batches = DevicePrefetcher(self.train_data, self.device)
+        else:
+            batches = self.train_data
         non_blocking = self.device.type == "cuda"
-        for source, targets in self.train_data:
-            source = source.to(self.device, non_blocking=non_blocking)
-            targets = targets.to(self.device, non_blocking=non_blocking)
+        for source, targets in batches:
+            if source.device != self.device:
+                source = source.to(self.device, non_blocking=non_blocking)
+                targets = targets.to(self.device, non_blocking=non_blocking)
             self._run_batch(source, targets)
             if self._profiler is not None:
                 self._profiler.step()

@@ -66,3 +66,16 @@ def test_dataloader_batch_shapes():
     dl = prepare_dataloader(ToyDataset(64), 32)
     x, y = next(iter(dl))
     assert x.shape == (32, 20) and y.shape == (32, 1)
+
+
+def test_device_prefetcher_cpu_passthrough():
+    import torch
+    from mi355x_ddp.data import DevicePrefetcher, ToyDataset, prepare_dataloader
+    ds = ToyDataset(64, seed=0)
+    loader = prepare_dataloader(ds, 8, shuffle=False)
+    pf = DevicePrefetcher(loader, torch.device("cpu"))
+    a = [(x.clone(), t.clone()) for x, t in loader]
+    b = [(x.clone(), t.clone()) for x, t in pf]
+    assert len(a) == len(b) == len(pf)
+    for (x1, t1), (x2, t2) in zip(a, b):
+        assert torch.equal(x1, x2) and torch.equal(t1, t2)
