@@ -67,7 +67,9 @@ class DeviceData:
                      else torch.Generator())
         # epoch shards are gathered on a SIDE stream one epoch ahead, so the
         # sort+gather overlaps the previous epoch's training kernels
-        self._side = torch.cuda.Stream() if self._cuda else None
+        # (MI355X_EPOCH_SIDE_STREAM=0 gathers inline on the main stream)
+        use_side = os.environ.get("MI355X_EPOCH_SIDE_STREAM", "1") != "0"
+        self._side = torch.cuda.Stream() if (self._cuda and use_side) else None
         self._pending = {}  # epoch -> (xs, ts, ready_event)
 
     def _gather(self, epoch):
@@ -80,7 +82,7 @@ class DeviceData:
         return self.X[shard].contiguous(), self.T[shard].contiguous()
 
     def _prep(self, epoch):
-        if not self._cuda:
+        if self._side is None:
             self._pending[epoch] = (*self._gather(epoch), None)
             return
         with torch.cuda.stream(self._side):
