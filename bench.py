@@ -65,10 +65,11 @@ class DeviceData:
         self._epoch = -1
         self._gen = (torch.Generator(device=device) if self._cuda
                      else torch.Generator())
-        # epoch shards are gathered on a SIDE stream one epoch ahead, so the
-        # sort+gather overlaps the previous epoch's training kernels
-        # (MI355X_EPOCH_SIDE_STREAM=0 gathers inline on the main stream)
-        use_side = os.environ.get("MI355X_EPOCH_SIDE_STREAM", "1") != "0"
+        # MI355X_EPOCH_SIDE_STREAM=1 gathers epoch shards on a side stream
+        # one epoch ahead; measured SLOWER than the inline default (A/B on
+        # one box: 9.9 vs 10.8 M samples/s) — the cross-stream allocation
+        # + event bookkeeping costs more than the ~25 us/epoch it hides.
+        use_side = os.environ.get("MI355X_EPOCH_SIDE_STREAM", "0") == "1"
         self._side = torch.cuda.Stream() if (self._cuda and use_side) else None
         self._pending = {}  # epoch -> (xs, ts, ready_event)
 
