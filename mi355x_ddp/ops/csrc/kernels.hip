@@ -1022,8 +1022,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
           if (row < B_) dy_s[row] = dy;
         }
     }
-    // single wave: LDS ops are program-ordered across lanes; the
-    // compiler's lgkmcnt waits order the dy_s write->read, no barrier
+    __syncthreads();  // dy_s visible to all lanes
 
     // ---- backward: dw_k = sum_i dY_i X[i,k] ----
     float av[BS];
@@ -1056,13 +1055,14 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     dbp = wave_sum(dbp);
     if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
 
-    // (ws/dy reads above are wave-ordered before the update below)
+    __syncthreads();  // all ws/dy reads done before the update
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
       if (q == 0 && k < K_) ws[k] = round_store<T>(ws[k] - lr * gacc[tk][0]);
     }
     if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
+    __syncthreads();
   }
 
   if (lane < K_) stf(&param[w_off + lane], ws[lane]);
