@@ -144,7 +144,7 @@ __global__ void k_linear_fwd_splitk(const T* __restrict__ X,
   const int q = lane >> 4;
   const int m = tile_m * 16 + r;
   const int n = tile_n * 16 + r;
-  const int kq = ((cdiv(K, 4) + 3) / 4) * 4;  // K-slice per wave (mult of 4)
+  const int kq = (((K + 3) / 4 + 3) / 4) * 4;  // K-slice per wave (mult of 4)
   const int k_lo = wave * kq, k_hi = min(K, (wave + 1) * kq);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
@@ -325,7 +325,7 @@ __global__ void k_linear_bwd_x_splitk(const T* __restrict__ dY,
   const int q = lane >> 4;
   const int m = tile_m * 16 + r;
   const int kc = tile_k * 16 + r;
-  const int nq = ((cdiv(N, 4) + 3) / 4) * 4;
+  const int nq = (((N + 3) / 4 + 3) / 4) * 4;
   const int j_lo = wave * nq, j_hi = min(N, (wave + 1) * nq);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
