@@ -72,6 +72,7 @@ class DeviceData:
         use_side = os.environ.get("MI355X_EPOCH_SIDE_STREAM", "0") == "1"
         self._side = torch.cuda.Stream() if (self._cuda and use_side) else None
         self._pending = {}  # epoch -> (xs, ts, ready_event)
+        self.bound_epoch = -1  # used by the shard-bound engine path
 
     def _gather(self, epoch):
         # epoch-seeded permutation generated ON DEVICE (identical on every
@@ -92,6 +93,20 @@ class DeviceData:
             ev.record(self._side)
         self._pending[epoch] = (xs, ts, ev)
 
+    def shard_for(self, epoch):
+        """The gathered epoch shard (xs, ts), for shard-bound engines —
+        no batch views are built on this path."""
+        if epoch not in self._pending:
+            self._prep(epoch)
+        xs, ts, ev = self._pending.pop(epoch)
+        if ev is not None:
+            main = torch.cuda.current_stream()
+            main.wait_event(ev)
+            xs.record_stream(main)
+            ts.record_stream(main)
+        self._prep(epoch + 1)
+        return xs, ts
+
     def batch_for(self, step):
         epoch, s = divmod(step, self.steps_per_epoch)
         if epoch != self._epoch:
@@ -107,9 +122,9 @@ class DeviceData:
                 xs.record_stream(main)
                 ts.record_stream(main)
             self._prep(epoch + 1)  # prefetch next epoch on the side stream
-            b = self.batch
-            self._views = [(xs[i * b:(i + 1) * b], ts[i * b:(i + 1) * b])
-                           for i in range(self.steps_per_epoch)]
+            # one C-side split call instead of 2*steps Python slicings
+            self._views = list(zip(torch.split(xs, self.batch),
+                                   torch.split(ts, self.batch)))
             self._epoch = epoch
         return self._views[s]
 
@@ -135,7 +150,7 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
             loss.backward()
             engine.finalize_backward()
             opt.step()
-        return step, noflush
+        return step, noflush, None
 
     if kind == "persistent" and comm is not None:
         kind = "fused"  # multi-step kernel is the world-1 in-kernel-SGD path
@@ -144,7 +159,7 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
     eng = cls(model, comm=comm, lr=lr, use_mse=True)
     if comm is not None:
         eng.reducer.broadcast_params(root=0)
-    return eng.step, getattr(eng, "flush", noflush)
+    return eng.step, getattr(eng, "flush", noflush), eng
 
 
 def main():
@@ -174,16 +189,36 @@ def main():
             torch.cuda.synchronize()
 
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    step_fn, flush_fn = build_engine(args.engine if use_cuda else "autograd",
-                                     comm, args.lr, device,
-                                     dtype if use_cuda else torch.float32)
+    step_fn, flush_fn, engine_obj = build_engine(
+        args.engine if use_cuda else "autograd", comm, args.lr, device,
+        dtype if use_cuda else torch.float32)
     data = DeviceData(args.dataset, rank, world, args.batch, device,
                       dtype=dtype if use_cuda else torch.float32)
 
+    # shard-bound fast path: bind each epoch's device-resident shard once,
+    # then drive the engine by batch INDEX (no per-step view construction)
+    eng_bind = getattr(engine_obj, "bind_shard", None)
+    eng_step_shard = getattr(engine_obj, "step_shard", None)
+    spe = data.steps_per_epoch
+
+    def run_steps(start, n):
+        if eng_bind is None:
+            for s in range(start, start + n):
+                x, t = data.batch_for(s)
+                step_fn(x, t)
+            return
+        cur = data.bound_epoch
+        step_shard = eng_step_shard
+        for s in range(start, start + n):
+            e, i = divmod(s, spe)
+            if e != cur:
+                xs, ts = data.shard_for(e)
+                eng_bind(xs, ts, args.batch)
+                data.bound_epoch = cur = e
+            step_shard(i)
+
     # -- warmup (untimed) -------------------------------------------------
-    for s in range(args.warmup):
-        x, t = data.batch_for(s)
-        step_fn(x, t)
+    run_steps(0, args.warmup)
     flush_fn()
     barrier()
 
@@ -191,9 +226,7 @@ def main():
     #    INSIDE the bracket — all K steps' work executes before the
     #    closing barrier+synchronize) ------------------------------------
     t0 = time.perf_counter()
-    for s in range(args.steps):
-        x, t = data.batch_for(args.warmup + s)
-        step_fn(x, t)
+    run_steps(args.warmup, args.steps)
     flush_fn()
     barrier()
     elapsed = time.perf_counter() - t0

@@ -90,53 +90,100 @@ class PersistentToyStep(ToyFusedStep):
         assert self.comm is None, \
             "PersistentToyStep is the world-1 path (in-kernel SGD)"
         self.max_defer = max_defer
-        self._x0 = None        # first batch of the pending run (keeps storage)
-        self._t0 = None
-        self._count = 0
-        self._nx = 0           # expected data_ptr of the next contiguous batch
-        self._nt = 0
-        self._bk = 0           # batch strides in bytes
-        self._bt = 0
-        self._shape = None
+        self._install_fast_path()
 
-    def step(self, x: torch.Tensor, t: torch.Tensor) -> None:
-        if self._count > 0:
-            # fast path: the next contiguous slice of the pending run
-            if (x.data_ptr() == self._nx and t.data_ptr() == self._nt
-                    and x.shape == self._shape):
-                self._count += 1
-                self._nx += self._bk
-                self._nt += self._bt
-                if self._count >= self.max_defer:
-                    self.flush()
+    def _install_fast_path(self) -> None:
+        """step/flush as closures over cell variables: the per-step cost is
+        two data_ptr() calls and a shape compare, no attribute traffic —
+        this loop runs a few hundred ns behind a kernel that takes ~1.3 us
+        per step, so host bookkeeping is the wall-clock bound."""
+        ext_mod = ops.ext() if ops.has_ext() else None
+        flat_param = self.flat_param
+        loss_out = self.loss_out
+        use_mse, w_off, b_off = self.use_mse, self.w_off, self.b_off
+        lr, max_defer = self.lr, self.max_defer
+        eager = ToyFusedStep.step.__get__(self)
+        x0 = t0 = None
+        shape = None
+        count = nx = nt = bk = bt = 0
+
+        def step(x: torch.Tensor, t: torch.Tensor) -> None:
+            nonlocal x0, t0, count, nx, nt, bk, bt, shape
+            if s_next != s_start:  # a shard-bound run is pending: order it
+                flush()
+            if count:
+                # fast path: the next contiguous slice of the pending run
+                if (x.data_ptr() == nx and t.data_ptr() == nt
+                        and x.shape == shape):
+                    count += 1
+                    nx += bk
+                    nt += bt
+                    if count >= max_defer:
+                        flush()
+                    return
+                flush()
+            if not (x.is_cuda and x.is_contiguous() and t.is_contiguous()):
+                eager(x, t)
                 return
-            self.flush()
-        if not (x.is_cuda and x.is_contiguous() and t.is_contiguous()):
-            super().step(x, t)
-            return
-        self._x0, self._t0, self._count = x, t, 1
-        self._shape = x.shape
-        self._bk = x.numel() * x.element_size()
-        self._bt = t.numel() * t.element_size()
-        self._nx = x.data_ptr() + self._bk
-        self._nt = t.data_ptr() + self._bt
+            x0, t0, count = x, t, 1
+            shape = x.shape
+            bk = x.numel() * x.element_size()
+            bt = t.numel() * t.element_size()
+            nx = x.data_ptr() + bk
+            nt = t.data_ptr() + bt
 
-    def flush(self) -> None:
-        if self._count == 0:
-            return
-        x0, t0, n = self._x0, self._t0, self._count
-        self._x0 = self._t0 = None
-        self._count = 0
-        B, K = x0.shape
-        if n == 1:
-            super().step(x0, t0)
-            return
-        xall = x0.as_strided((n * B, K), (K, 1))
-        tall = t0.as_strided((n * B,) + t0.shape[1:],
-                             (t0.stride(0),) + t0.stride()[1:])
-        ops.ext().toy_multistep(xall, tall, self.flat_param, self.loss_out,
-                                self.use_mse, self.w_off, self.b_off,
-                                self.lr, B)
+        # shard-bound path: bind the epoch shard once, then step by batch
+        # INDEX — no per-step tensor-view construction at all. Used by
+        # bench.py's device-resident data path; step(x, t) remains the
+        # generic API.
+        shard_x = shard_t = None
+        sbatch = 0
+        s_start = s_next = 0  # pending run = batch indices [s_start, s_next)
+
+        def bind_shard(xs: torch.Tensor, ts: torch.Tensor, batch: int) -> None:
+            nonlocal shard_x, shard_t, sbatch, s_start, s_next
+            flush()
+            assert xs.is_cuda and xs.is_contiguous() and ts.is_contiguous()
+            shard_x, shard_t, sbatch = xs, ts, batch
+            s_start = s_next = 0
+
+        def step_shard(i: int) -> None:
+            nonlocal s_start, s_next
+            if i == s_next:
+                s_next += 1
+                if s_next - s_start >= max_defer:
+                    flush()
+                return
+            flush()
+            s_start, s_next = i, i + 1
+
+        def flush() -> None:
+            nonlocal x0, t0, count, s_start, s_next
+            if count:
+                xx, tt, n = x0, t0, count
+                x0 = t0 = None
+                count = 0
+                if n == 1:
+                    eager(xx, tt)
+                else:
+                    B, K = xx.shape
+                    xall = xx.as_strided((n * B, K), (K, 1))
+                    tall = tt.as_strided((n * B,) + tt.shape[1:],
+                                         (tt.stride(0),) + tt.stride()[1:])
+                    ext_mod.toy_multistep(xall, tall, flat_param, loss_out,
+                                          use_mse, w_off, b_off, lr, B)
+            if s_next > s_start:
+                lo, n = s_start * sbatch, (s_next - s_start) * sbatch
+                s_start = s_next
+                xx = shard_x[lo:lo + n]
+                tt = shard_t[lo:lo + n]
+                ext_mod.toy_multistep(xx, tt, flat_param, loss_out,
+                                      use_mse, w_off, b_off, lr, sbatch)
+
+        self.step = step
+        self.flush = flush
+        self.bind_shard = bind_shard
+        self.step_shard = step_shard
 
 
 class GraphedToyStep(ToyFusedStep):

@@ -149,3 +149,52 @@ def test_bench_single_gpu_json():
     r = json.loads(line)
     assert r["n_gpus"] == 1 and r["steps"] == 50
     assert r["value"] > 0 and r["dtype"] == "fp32"
+
+
+def test_shard_bound_matches_single_step():
+    from mi355x_ddp.engine import PersistentToyStep
+    from mi355x_ddp.models import toy_model
+    w_ref, b_ref = _train_fused_dtype(torch.float32)
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(DEV)
+    eng = PersistentToyStep(model, comm=None, lr=LR, use_mse=True)
+    X, T = _data()
+    Xf = X.reshape(STEPS * 32, 20).contiguous()
+    Tf = T.reshape(STEPS * 32, 1).contiguous()
+    eng.bind_shard(Xf, Tf, 32)
+    for s in range(STEPS):
+        eng.step_shard(s)
+    eng.flush()
+    torch.cuda.synchronize()
+    assert torch.equal(model.weight.detach().cpu(), w_ref)
+    assert torch.equal(model.bias.detach().cpu(), b_ref)
+
+
+def test_shard_bound_non_sequential_indices():
+    from mi355x_ddp.engine import PersistentToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    order = [0, 1, 2, 7, 8, 3, 4, 19]
+    X, T = _data()
+    Xf = X.reshape(STEPS * 32, 20).contiguous()
+    Tf = T.reshape(STEPS * 32, 1).contiguous()
+
+    torch.manual_seed(7)
+    ref_model = toy_model(20, 1).to(DEV)
+    ref_eng = ToyFusedStep(ref_model, comm=None, lr=LR, use_mse=True)
+    for i in order:
+        ref_eng.step(Xf[i * 32:(i + 1) * 32].contiguous(),
+                     Tf[i * 32:(i + 1) * 32].contiguous())
+    torch.cuda.synchronize()
+
+    torch.manual_seed(7)
+    model = toy_model(20, 1).to(DEV)
+    eng = PersistentToyStep(model, comm=None, lr=LR, use_mse=True)
+    eng.bind_shard(Xf, Tf, 32)
+    for i in order:
+        eng.step_shard(i)
+    eng.flush()
+    torch.cuda.synchronize()
+    assert torch.equal(model.weight.detach().cpu(),
+                       ref_model.weight.detach().cpu())
+    assert torch.equal(model.bias.detach().cpu(),
+                       ref_model.bias.detach().cpu())
