@@ -1165,14 +1165,15 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     dbp = wave_sum(dbp);
     if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
 
-    __syncthreads();  // all ws/dy reads done before the update
+    // no barrier needed here: the dy_s barrier above already ordered this
+    // step's ws READS (forward) before these writes
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
       if (q == 0 && k < K_) ws[k] = round_store<T>(ws[k] - lr * gacc[tk][0]);
     }
     if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
-    __syncthreads();
+    __syncthreads();  // ws update visible before next iteration's forward
   }
 
   if (lane < K_) stf(&param[w_off + lane], ws[lane]);
