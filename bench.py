@@ -73,8 +73,17 @@ class DeviceData:
         self._side = torch.cuda.Stream() if (self._cuda and use_side) else None
         self._pending = {}  # epoch -> (xs, ts, ready_event)
         self.bound_epoch = -1  # used by the shard-bound engine path
+        self._fused_shard = os.environ.get("MI355X_SHUFFLE", "") != "randperm"
 
     def _gather(self, epoch):
+        if self._cuda and self._fused_shard:
+            # ONE kernel: epoch-seeded bijective permutation computed
+            # inline per row + linear copy (replaces randperm's radix sort
+            # + two index_selects; MI355X_SHUFFLE=randperm for the old path)
+            from mi355x_ddp import ops
+            xs, ts = ops.ext().epoch_shard(self.X, self.T, 1000 + epoch,
+                                           self.rank, self.world)
+            return xs, ts
         # epoch-seeded permutation generated ON DEVICE (identical on every
         # rank for a given epoch; no host round trip)
         self._gen.manual_seed(1000 + epoch)

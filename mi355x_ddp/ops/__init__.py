@@ -156,6 +156,34 @@ def sgd_flat_(param_flat: torch.Tensor, grad_flat: torch.Tensor, lr: float,
             grad_flat.zero_()
 
 
+def perm_index(seed: int, p: int, n: int) -> int:
+    """Python mirror of the kernel's seeded bijection on [0, n)
+    (kernels.hip `mix_bijection`/`shard_consts`): source row for shard
+    position p at epoch seed `seed`. Used by tests and the CPU data path."""
+    M = 0xFFFFFFFF
+    z = (seed * 0x9E3779B9 + 0x7F4A7C15) & M
+    z ^= z >> 15
+    z = (z * 0x2C1B3C6D) & M
+    z ^= z >> 12
+    c0, m0 = z, ((z >> 8) | 1) & M
+    z = (z * 0x297A2D39 + 0x68E31DA4) & M
+    z ^= z >> 16
+    c1, m1 = z, ((z >> 7) | 1) & M
+    k_mask = 1
+    while k_mask + 1 < n:
+        k_mask = (k_mask << 1) | 1
+    x = p
+    while True:
+        x = (x + c0) & k_mask
+        x = (x * m0) & k_mask
+        x ^= x >> 3
+        x = (x + c1) & k_mask
+        x = (x * m1) & k_mask
+        x ^= x >> 5
+        if x < n:
+            return x
+
+
 def cpu_linear_bwd_weight(x, dy, dw, db, accumulate=False):
     """CPU reference for the bwd-weight kernel (used in numerics tests)."""
     w_new = dy.t() @ x

@@ -725,6 +725,83 @@ void unflatten_from(torch::Tensor bucket, torch::Tensor plan, int64_t total_bloc
 }
 
 // ---------------------------------------------------------------------------
+// Epoch shard gather with an IN-KERNEL seeded permutation: dest row i of
+// this rank's shard copies source row perm(seed, rank + i*world) of the
+// device-resident dataset. perm is a seeded bijection on [0, n) (mix of
+// add/odd-multiply/xorshift rounds on ceil-log2 bits with cycle-walking),
+// so one linear copy pass replaces torch.randperm's radix sort + two
+// index_select launches (~30 us -> ~3 us per epoch). Python mirror:
+// mi355x_ddp.ops.perm_index (tested for permutation property + parity).
+// ---------------------------------------------------------------------------
+__device__ __host__ __forceinline__ uint32_t mix_bijection(
+    uint32_t x, uint32_t k_mask, uint32_t n, uint32_t c0, uint32_t m0,
+    uint32_t c1, uint32_t m1) {
+  // each op is bijective on (k_mask+1)-space; cycle-walk back into [0, n)
+  do {
+    x = (x + c0) & k_mask;
+    x = (x * m0) & k_mask;       // m0 odd
+    x ^= x >> 3;
+    x = (x + c1) & k_mask;
+    x = (x * m1) & k_mask;       // m1 odd
+    x ^= x >> 5;
+  } while (x >= n);
+  return x;
+}
+
+__device__ __host__ __forceinline__ void shard_consts(
+    uint32_t seed, uint32_t* c0, uint32_t* m0, uint32_t* c1, uint32_t* m1) {
+  uint32_t z = seed * 0x9E3779B9u + 0x7F4A7C15u;
+  z ^= z >> 15; z *= 0x2C1B3C6Du; z ^= z >> 12;
+  *c0 = z;
+  *m0 = (z >> 8) | 1u;
+  z = z * 0x297A2D39u + 0x68E31DA4u; z ^= z >> 16;
+  *c1 = z;
+  *m1 = (z >> 7) | 1u;
+}
+
+template <typename T>
+__global__ void k_epoch_shard(const T* __restrict__ X, const T* __restrict__ Tg,
+                              T* __restrict__ xs, T* __restrict__ ts,
+                              int n, int K, int per_rank, int rank, int world,
+                              uint32_t k_mask, uint32_t c0, uint32_t m0,
+                              uint32_t c1, uint32_t m1) {
+  const int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  const int cols = K + 1;
+  if (tid >= per_rank * cols) return;
+  const int i = tid / cols;
+  const int c = tid - i * cols;
+  const uint32_t p = (uint32_t)(rank + (size_t)i * world);
+  const uint32_t src = mix_bijection(p, k_mask, (uint32_t)n, c0, m0, c1, m1);
+  if (c < K) xs[(size_t)i * K + c] = X[(size_t)src * K + c];
+  else       ts[i] = Tg[src];
+}
+
+std::vector<torch::Tensor> epoch_shard(torch::Tensor X, torch::Tensor Tg,
+                                       int64_t seed, int64_t rank,
+                                       int64_t world) {
+  TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Tg.is_contiguous());
+  const int n = (int)X.size(0), K = (int)X.size(1);
+  TORCH_CHECK(Tg.size(0) == n && Tg.size(1) == 1, "targets must be [n,1]");
+  const int per_rank = n / (int)world;
+  auto xs = at::empty({per_rank, K}, X.options());
+  auto ts = at::empty({per_rank, 1}, Tg.options());
+  uint32_t k_mask = 1;
+  while ((int64_t)k_mask + 1 < n) k_mask = (k_mask << 1) | 1u;
+  uint32_t c0, m0, c1, m1;
+  shard_consts((uint32_t)seed, &c0, &m0, &c1, &m1);
+  const int total = per_rank * (K + 1);
+  DISPATCH_F32_BF16(X.scalar_type(), "epoch_shard", {
+    hipLaunchKernelGGL((k_epoch_shard<scalar_t>), dim3(cdiv(total, 256)),
+                       dim3(256), 0, cur_stream(), cdptr<scalar_t>(X),
+                       cdptr<scalar_t>(Tg), dptr<scalar_t>(xs),
+                       dptr<scalar_t>(ts), n, K, per_rank, (int)rank,
+                       (int)world, k_mask, c0, m0, c1, m1);
+  });
+  HIP_OK(hipGetLastError());
+  return {xs, ts};
+}
+
+// ---------------------------------------------------------------------------
 // Fused toy training step (fwd + loss-grad + bwd [+ SGD] in ONE kernel):
 // the reference hot loop single_gpu.py:21-26 for model = Linear(K,1).
 // Single workgroup, 64 threads (one wave): at 84 B of gradients the step is

@@ -62,6 +62,12 @@ void toy_fused_fwd_bwd(torch::Tensor x, torch::Tensor t,
                        torch::Tensor loss_out, bool use_mse,
                        int64_t w_off, int64_t b_off, double lr);
 
+// Epoch shard gather with an in-kernel seeded permutation (see
+// kernels.hip): returns (xs [n/world, K], ts [n/world, 1]) for this rank.
+std::vector<torch::Tensor> epoch_shard(torch::Tensor X, torch::Tensor Tg,
+                                       int64_t seed, int64_t rank,
+                                       int64_t world);
+
 // Multi-step persistent toy trainer (world-1): x is [S*batch, K] of S
 // consecutive batches; runs S full fwd+loss+bwd+SGD steps in ONE launch
 // with the weights resident in LDS (bitwise-identical per-step arithmetic

@@ -223,3 +223,33 @@ def test_single_gpu_trainer_runs(tmp_chdir):
     tr.train(2)
     import os
     assert os.path.exists("checkpoint.pt")
+
+
+def test_epoch_shard_matches_python_mirror():
+    # kernel permutation == ops.perm_index mirror, and the world's shards
+    # together form an exact permutation of the dataset rows
+    n, K, world = 2048, 20, 4
+    g = torch.Generator().manual_seed(5)
+    X = torch.rand(n, K, generator=g).to(DEV)
+    T = torch.rand(n, 1, generator=g).to(DEV)
+    seed = 1007
+    seen = []
+    per = n // world
+    for rank in range(world):
+        xs, ts = ops.ext().epoch_shard(X, T, seed, rank, world)
+        idx = [ops.perm_index(seed, rank + i * world, n) for i in range(per)]
+        seen += idx
+        ref_x = X[torch.tensor(idx, device=DEV)]
+        ref_t = T[torch.tensor(idx, device=DEV)]
+        assert torch.equal(xs, ref_x)
+        assert torch.equal(ts, ref_t)
+    assert sorted(seen) == list(range(n))  # exact permutation, no repeats
+
+
+def test_epoch_shard_different_epochs_differ():
+    n, K = 256, 20
+    X = torch.rand(n, K).to(DEV)
+    T = torch.rand(n, 1).to(DEV)
+    a, _ = ops.ext().epoch_shard(X, T, 1, 0, 1)
+    b, _ = ops.ext().epoch_shard(X, T, 2, 0, 1)
+    assert not torch.equal(a, b)
