@@ -1125,21 +1125,18 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
   // All loads are UNCONDITIONAL with clamped addresses + value selects:
   // ternary-guarded loads compile to exec-masked branch-per-element code
   // that serializes the burst (seen in the r01b ISA), selects do not.
-  struct RegSet {
-    float fA[MT][KS];  // fwd A: X[tm*16+r][4*kk+q]
-    float bB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
-    float tR[MT][4];   // targets for rows tm*16+q*4+i
-  };
-  RegSet ra, rb;  // alternating sets: true software pipeline, no copies
+  float cfA[MT][KS];  // fwd A: X[tm*16+r][4*kk+q]
+  float cbB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
+  float ctR[MT][4];   // targets for rows tm*16+q*4+i
 
-  auto prefetch = [&](RegSet& R, int s) {
+  auto prefetch = [&](int s) {
     const T* Xs = X + (size_t)s * (B_ * K_);
     const T* Ts = Tg + (size_t)s * B_;
     // Loads are raw (address-clamped, no value select): out-of-range
     // lanes load in-bounds garbage whose products are either multiplied
     // by a zeroed LDS-side operand (wv/av) or discarded by the guarded
-    // ws/dy writes — so no per-load mask, and no vmcnt wait until the
-    // NEXT iteration's first MFMA use of this register set.
+    // ws/dy writes — so no per-load mask, and therefore no vmcnt wait
+    // until the next iteration's first MFMA use.
 #pragma unroll
     for (int tm = 0; tm < MT; ++tm) {
       const int m = tm * 16 + r;
@@ -1148,7 +1145,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       for (int kk = 0; kk < KS; ++kk) {
         const int k = 4 * kk + q;
         const int kc = (k < K_) ? k : K_ - 1;
-        R.fA[tm][kk] = ldf(&Xs[mc * K_ + kc]);
+        cfA[tm][kk] = ldf(&Xs[mc * K_ + kc]);
       }
     }
 #pragma unroll
@@ -1159,7 +1156,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       for (int ii = 0; ii < BS; ++ii) {
         const int i = 4 * ii + q;
         const int ic = (i < B_) ? i : B_ - 1;
-        R.bB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
+        cbB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
       }
     }
 #pragma unroll
@@ -1167,16 +1164,13 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int row = tm * 16 + q * 4 + i;
-        R.tR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
+        ctR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
       }
   };
 
+  prefetch(0);
   float loss_last = 0.f;
-  auto body = [&](RegSet& cur, RegSet& nxt, int s) {
-    // issue the NEXT step's loads first: they retire behind this whole
-    // step's math and are only waited on at the next body's forward
-    if (s + 1 < S) prefetch(nxt, s + 1);
-
+  for (int s = 0; s < S; ++s) {
     // batch the w reads (LDS; one unconditional read + select per kk)
     float wv[KS];
 #pragma unroll
@@ -1195,7 +1189,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     for (int kk = 0; kk < KS; ++kk)
 #pragma unroll
       for (int tm = 0; tm < MT; ++tm)
-        acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(cur.fA[tm][kk], wv[kk],
+        acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(cfA[tm][kk], wv[kk],
                                                        acc[tm], 0, 0, 0);
 
     // ---- loss grad (rows live on r==0 lanes) + dY exchange ----
@@ -1208,7 +1202,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
           const int row = tm * 16 + q * 4 + i;
           float dy = 0.f;
           if (row < B_ && use_mse) {
-            const float d = acc[tm][i] + bterm - cur.tR[tm][i];
+            const float d = acc[tm][i] + bterm - ctR[tm][i];
             loss_acc += d * d;
             dy = d * inv2B;
           }
@@ -1232,8 +1226,13 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     for (int ii = 0; ii < BS; ++ii)
 #pragma unroll
       for (int tk = 0; tk < KT; ++tk)
-        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], cur.bB[tk][ii],
+        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], cbB[tk][ii],
                                                         gacc[tk], 0, 0, 0);
+
+    // cur regs are dead from here: load next step's operands straight into
+    // them — the vmcnt wait attaches to their first use (next iteration's
+    // forward MFMA), shadowed by the update/barrier below
+    if (s + 1 < S) prefetch(s + 1);
 
     // db: same lane-strided dy_s reduction order as the single-step
     // kernel (keeps the multi-step history bitwise-identical)
@@ -1252,15 +1251,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     }
     if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
     __syncthreads();  // ws update visible before next iteration's forward
-  };
-
-  prefetch(ra, 0);
-  int s = 0;
-  for (; s + 2 <= S; s += 2) {
-    body(ra, rb, s);
-    body(rb, ra, s + 1);
   }
-  if (s < S) body(ra, rb, s);
 
   if (lane < K_) stf(&param[w_off + lane], ws[lane]);
   if (lane == K_) stf(&param[b_off], ws[32]);
