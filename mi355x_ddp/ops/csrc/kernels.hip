@@ -1129,17 +1129,14 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
   float cbB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
   float ctR[MT][4];   // targets for rows tm*16+q*4+i
 
-  // Loads are raw (address-clamped, no value select): out-of-range lanes
-  // load in-bounds garbage whose products are either multiplied by a
-  // zeroed LDS-side operand (wv/av) or discarded by the guarded ws/dy
-  // writes — so no per-load mask, and no vmcnt wait until the next
-  // iteration's first MFMA use. The prefetch is SPLIT: each register
-  // group is reloaded immediately after its last use in the step, giving
-  // every load at least half a step of latency shadow with no extra
-  // registers.
-  auto prefetch_fwd = [&](int s) {  // cfA + ctR: dead after forward/dy
+  auto prefetch = [&](int s) {
     const T* Xs = X + (size_t)s * (B_ * K_);
     const T* Ts = Tg + (size_t)s * B_;
+    // Loads are raw (address-clamped, no value select): out-of-range
+    // lanes load in-bounds garbage whose products are either multiplied
+    // by a zeroed LDS-side operand (wv/av) or discarded by the guarded
+    // ws/dy writes — so no per-load mask, and therefore no vmcnt wait
+    // until the next iteration's first MFMA use.
 #pragma unroll
     for (int tm = 0; tm < MT; ++tm) {
       const int m = tm * 16 + r;
@@ -1152,16 +1149,6 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       }
     }
 #pragma unroll
-    for (int tm = 0; tm < MT; ++tm)
-#pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row = tm * 16 + q * 4 + i;
-        ctR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
-      }
-  };
-  auto prefetch_bwd = [&](int s) {  // cbB: dead after the backward MFMAs
-    const T* Xs = X + (size_t)s * (B_ * K_);
-#pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
       const int kc = (k < K_) ? k : K_ - 1;
@@ -1172,10 +1159,16 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
         cbB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
       }
     }
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm)
+#pragma unroll
+      for (int i = 0; i < 4; ++i) {
+        const int row = tm * 16 + q * 4 + i;
+        ctR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
+      }
   };
 
-  prefetch_fwd(0);
-  prefetch_bwd(0);
+  prefetch(0);
   float loss_last = 0.f;
   for (int s = 0; s < S; ++s) {
     // batch the w reads (LDS; one unconditional read + select per kk)
@@ -1216,7 +1209,6 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
           if (row < B_) dy_s[row] = dy;
         }
     }
-    if (s + 1 < S) prefetch_fwd(s + 1);  // cfA/ctR now dead
     __syncthreads();  // dy_s visible to all lanes
 
     // ---- backward: dw_k = sum_i dY_i X[i,k] ----
@@ -1237,7 +1229,10 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
         gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], cbB[tk][ii],
                                                         gacc[tk], 0, 0, 0);
 
-    if (s + 1 < S) prefetch_bwd(s + 1);  // cbB now dead
+    // cur regs are dead from here: load next step's operands straight into
+    // them — the vmcnt wait attaches to their first use (next iteration's
+    // forward MFMA), shadowed by the update/barrier below
+    if (s + 1 < S) prefetch(s + 1);
 
     // db: same lane-strided dy_s reduction order as the single-step
     // kernel (keeps the multi-step history bitwise-identical)
