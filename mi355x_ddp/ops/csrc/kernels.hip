@@ -1101,6 +1101,147 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
 // and the backward MFMA. Divisions are hoisted to one reciprocal
 // (2/B is exact for the reference batch 32, so dy is bitwise-unchanged).
 // ---------------------------------------------------------------------------
+// bf16 wide-MFMA variant: the whole K (<=32) contraction of the forward and
+// the whole B (=32) contraction of the backward are each ONE
+// v_mfma_f32_16x16x32_bf16 per tile (4 MFMAs/step vs 26 in the f32 path),
+// with operands kept as raw bf16x8 registers — no per-element up-convert.
+// dY is rounded to bf16 before the backward MFMA (true bf16 compute), so
+// this path matches the f32-MFMA path to bf16 accuracy, not bitwise
+// (tests/test_engine_gpu.py uses allclose for bf16).
+// Operand map (as k_gemm_bf16): A[l&15][(l>>4)*8+j]; B[(l>>4)*8+j][l&15];
+// D col=l&15, row=(l>>4)*4+reg.
+// ---------------------------------------------------------------------------
+template <int B_, int K_>
+__global__ void __launch_bounds__(64, 1)
+k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
+                      const __hip_bfloat16* __restrict__ Tg,
+                      __hip_bfloat16* __restrict__ param,
+                      float* __restrict__ loss_out,
+                      int S, int use_mse, int w_off, int b_off, float lr) {
+  static_assert(B_ == 32 && K_ <= 32, "bf16 wide path: B=32, K<=32");
+  constexpr int MT = 2;                // fwd 16-row tiles
+  constexpr int KT = (K_ + 15) / 16;   // bwd 16-col tiles
+  const int lane = threadIdx.x;
+  const int r = lane & 15, q = lane >> 4;
+  __shared__ float ws[33];
+  __shared__ float dy_s[32];
+
+  if (lane < K_) ws[lane] = ldf(&param[w_off + lane]);
+  if (lane == K_) ws[32] = ldf(&param[b_off]);
+  const float inv2B = 2.f / (float)B_;
+
+  bf16x8 cfa[MT];      // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
+  bf16x8 cbb[KT];      // bwd B: X[q*8+j][tk*16+r] (raw)
+  float ctR[MT][4];    // targets for rows tm*16+q*4+i
+
+  auto prefetch = [&](int s) {
+    const __hip_bfloat16* Xs = X + (size_t)s * (B_ * K_);
+    const __hip_bfloat16* Ts = Tg + (size_t)s * B_;
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) {
+      const int m = tm * 16 + r;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int k = q * 8 + j;
+        const int kc = (k < K_) ? k : K_ - 1;
+        cfa[tm][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)m * K_ + kc]);
+      }
+    }
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      const int kc = (k < K_) ? k : K_ - 1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int i = q * 8 + j;  // < 32 == B_
+        cbb[tk][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)i * K_ + kc]);
+      }
+    }
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm)
+#pragma unroll
+      for (int i = 0; i < 4; ++i)
+        ctR[tm][i] = ldf(&Ts[tm * 16 + q * 4 + i]);
+  };
+
+  prefetch(0);
+  float loss_last = 0.f;
+  for (int s = 0; s < S; ++s) {
+    // forward B-operand: w as bf16 (stored values round-trip bf16 exactly),
+    // zero-padded for k >= K_ so raw garbage in A contributes nothing
+    bf16x8 b8{};
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int k = q * 8 + j;
+      const float v = ws[(k < K_) ? k : K_ - 1];
+      b8[j] = (r == 0 && k < K_) ? (__bf16)v : (__bf16)0.f;
+    }
+    const float bterm = ws[32];
+
+    f32x4 acc[MT];
+#pragma unroll
+    for (int tm = 0; tm < MT; ++tm) {
+      acc[tm] = {0.f, 0.f, 0.f, 0.f};
+      acc[tm] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(cfa[tm], b8,
+                                                        acc[tm], 0, 0, 0);
+    }
+
+    float loss_acc = 0.f;
+    if (r == 0) {
+#pragma unroll
+      for (int tm = 0; tm < MT; ++tm)
+#pragma unroll
+        for (int i = 0; i < 4; ++i) {
+          const int row = tm * 16 + q * 4 + i;
+          float dy = 0.f;
+          if (use_mse) {
+            const float d = acc[tm][i] + bterm - ctR[tm][i];
+            loss_acc += d * d;
+            dy = d * inv2B;
+          }
+          dy_s[row] = dy;
+        }
+    }
+    __syncthreads();  // dy_s visible to all lanes
+
+    // backward A-operand: dY rounded to bf16 (j-th element = dy[q*8+j])
+    bf16x8 a8{};
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const float v = dy_s[q * 8 + j];
+      a8[j] = (r == 0) ? (__bf16)v : (__bf16)0.f;
+    }
+    f32x4 gacc[KT];
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      gacc[tk] = {0.f, 0.f, 0.f, 0.f};
+      gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a8, cbb[tk],
+                                                         gacc[tk], 0, 0, 0);
+    }
+
+    if (s + 1 < S) prefetch(s + 1);  // operand regs dead from here
+
+    float dbp = 0.f;
+#pragma unroll
+    for (int i = lane; i < B_; i += 64) dbp += dy_s[i];
+    dbp = wave_sum(dbp);
+    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
+
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      if (q == 0 && k < K_)
+        ws[k] = round_store<__hip_bfloat16>(ws[k] - lr * gacc[tk][0]);
+    }
+    if (lane == 0) ws[32] = round_store<__hip_bfloat16>(bterm - lr * dbp);
+    __syncthreads();  // ws update visible before next iteration's forward
+  }
+
+  if (lane < K_) stf(&param[w_off + lane], ws[lane]);
+  if (lane == K_) stf(&param[b_off], ws[32]);
+  if (lane == 0 && loss_out) *loss_out = use_mse ? loss_last / (float)B_ : 0.f;
+}
+
 template <typename T, int B_, int K_>
 __global__ void __launch_bounds__(64, 1)
 k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
@@ -1267,9 +1408,16 @@ static void launch_toy_multistep(const torch::Tensor& x, const torch::Tensor& t,
   const T* tp = cdptr<T>(t);
   T* pp = dptr<T>(param_flat);
   if (B == 32 && K == 20) {  // the reference shape: compile-time fast path
-    hipLaunchKernelGGL((k_toy_multistep_spec<T, 32, 20>), dim3(1), dim3(64),
-                       0, cur_stream(), xp, tp, pp, lossp, S,
-                       use_mse ? 1 : 0, w_off, b_off, lr);
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      // native bf16 MFMA (16x16x32): whole contraction in one MFMA/tile
+      hipLaunchKernelGGL((k_toy_multistep_bf16w<32, 20>), dim3(1), dim3(64),
+                         0, cur_stream(), xp, tp, pp, lossp, S,
+                         use_mse ? 1 : 0, w_off, b_off, lr);
+    } else {
+      hipLaunchKernelGGL((k_toy_multistep_spec<T, 32, 20>), dim3(1), dim3(64),
+                         0, cur_stream(), xp, tp, pp, lossp, S,
+                         use_mse ? 1 : 0, w_off, b_off, lr);
+    }
     return;
   }
   auto go = [&](auto mt, auto kt) {

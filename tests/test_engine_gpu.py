@@ -113,15 +113,24 @@ def _train_fused_dtype(dtype):
             model.bias.detach().float().cpu().clone())
 
 
-@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
-def test_persistent_multistep_bitwise_matches_single_step(dtype):
-    # S deferred steps in ONE kernel == S single-step launches, bitwise:
-    # the multi-step kernel uses identical per-step arithmetic (including
-    # bf16 round-trip of the stored params between steps).
-    w_ref, b_ref = _train_fused_dtype(dtype)
-    w, b = _train_persistent(dtype=dtype)
+def test_persistent_multistep_bitwise_matches_single_step():
+    # S deferred steps in ONE kernel == S single-step launches, BITWISE:
+    # the f32 multi-step kernel uses identical per-step arithmetic.
+    w_ref, b_ref = _train_fused_dtype(torch.float32)
+    w, b = _train_persistent(dtype=torch.float32)
     assert torch.equal(w, w_ref), (w - w_ref).abs().max()
     assert torch.equal(b, b_ref)
+
+
+def test_persistent_multistep_bf16_wide_mfma_close():
+    # the bf16 multi-step path computes on v_mfma_f32_16x16x32_bf16 with dY
+    # rounded to bf16 (true bf16 pipeline) — matches the f32-MFMA
+    # single-step path to bf16 accuracy, not bitwise
+    w_ref, b_ref = _train_fused_dtype(torch.bfloat16)
+    w, b = _train_persistent(dtype=torch.bfloat16)
+    assert torch.allclose(w, w_ref, atol=5e-3, rtol=5e-2), \
+        (w - w_ref).abs().max()
+    assert torch.allclose(b, b_ref, atol=5e-3, rtol=5e-2)
 
 
 def test_persistent_fallback_scattered_batches():
