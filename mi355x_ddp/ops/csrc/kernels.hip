@@ -888,13 +888,19 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
     f32x4 acc[KT];
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) acc[tk] = {0.f, 0.f, 0.f, 0.f};
+    // db rides a FREE output column: when K is not a multiple of 16,
+    // column K of the bwd tile multiplies dY against a constant-1 operand,
+    // so the bias gradient falls out of the same MFMA chain — no LDS
+    // re-read, no wave_sum of 6 dependent cross-lane shuffles per step.
+    const bool mfma_db = (K & 15) != 0;
     for (int i0 = 0; i0 < B; i0 += 4) {
       const int i = i0 + q;
       const float a = (r == 0 && i < B) ? dy_s[i] : 0.f;  // A[j=0][i]
 #pragma unroll
       for (int tk = 0; tk < KT; ++tk) {
         const int k = tk * 16 + r;
-        const float b = (i < B && k < K) ? xs[i * K + k] : 0.f;
+        const float b = (i < B && k < K) ? xs[i * K + k]
+                       : (i < B && k == K && mfma_db) ? 1.f : 0.f;
         acc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tk], 0, 0, 0);
       }
     }
@@ -905,19 +911,23 @@ __global__ void k_toy_fused(const T* __restrict__ X, const T* __restrict__ Tg,
       if (q == 0 && k < K) {
         if (lr > 0.f) stf(&param[w_off + k], ws[k] - lr * acc[tk][0]);
         else stf(&grad[w_off + k], acc[tk][0]);
+      } else if (q == 0 && k == K && mfma_db) {
+        if (lr > 0.f) stf(&param[b_off], bterm - lr * acc[tk][0]);
+        else stf(&grad[b_off], acc[tk][0]);
+      }
+    }
+    if (!mfma_db) {
+      float dbp = 0.f;
+      for (int i = lane; i < B; i += 64) dbp += dy_s[i];
+      dbp = wave_sum(dbp);
+      if (lane == 0) {
+        if (lr > 0.f) stf(&param[b_off], bterm - lr * dbp);
+        else stf(&grad[b_off], dbp);
       }
     }
   }
-  // db + loss reduce
-  float dbp = 0.f;
-  for (int i = lane; i < B; i += 64) dbp += dy_s[i];
-  dbp = wave_sum(dbp);
-  if (use_mse) loss_acc = wave_sum(loss_acc);
-  if (lane == 0) {
-    if (lr > 0.f) stf(&param[b_off], bterm - lr * dbp);
-    else stf(&grad[b_off], dbp);
-    if (loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
-  }
+  if (use_mse && loss_out) loss_acc = wave_sum(loss_acc);
+  if (lane == 0 && loss_out) *loss_out = use_mse ? loss_acc / (float)B : 0.f;
 }
 
 // ---------------------------------------------------------------------------
@@ -1054,6 +1064,7 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
     __syncthreads();  // dy_s visible to all lanes before the bwd MFMAs
 
     // ---- backward + in-LDS SGD update ----
+    const bool mfma_db = (K & 15) != 0;  // free db column (as k_toy_fused)
     {
       f32x4 acc[KT];
 #pragma unroll
@@ -1064,7 +1075,8 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
 #pragma unroll
         for (int tk = 0; tk < KT; ++tk) {
           const int k = tk * 16 + r;
-          const float b = (i < B && k < K) ? xs[i * K + k] : 0.f;
+          const float b = (i < B && k < K) ? xs[i * K + k]
+                         : (i < B && k == K && mfma_db) ? 1.f : 0.f;
           acc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc[tk], 0, 0, 0);
         }
       }
@@ -1074,13 +1086,17 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
         const int k = tk * 16 + r;
         if (q == 0 && k < K)
           ws[k] = round_store<T>(ws[k] - lr * acc[tk][0]);
+        else if (q == 0 && k == K && mfma_db)
+          ws[32] = round_store<T>(bterm - lr * acc[tk][0]);
       }
     }
-    float dbp = 0.f;
-    for (int i = lane; i < B; i += 64) dbp += dy_s[i];
-    dbp = wave_sum(dbp);
-    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
-    if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
+    if (!mfma_db) {
+      float dbp = 0.f;
+      for (int i = lane; i < B; i += 64) dbp += dy_s[i];
+      dbp = wave_sum(dbp);
+      if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
+    }
+    if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
     __syncthreads();
   }
 
@@ -1212,28 +1228,32 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
       a8[j] = (r == 0) ? (__bf16)v : (__bf16)0.f;
     }
     f32x4 gacc[KT];
+    static_assert(K_ % 16 != 0, "free db column requires K_ % 16 != 0");
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
+      // db rides output column K_: constant-1 B operand on that lane
+      bf16x8 bb = cbb[tk];
+      if (tk * 16 + r == K_) {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) bb[j] = (__bf16)1.f;
+      }
       gacc[tk] = {0.f, 0.f, 0.f, 0.f};
-      gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a8, cbb[tk],
+      gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a8, bb,
                                                          gacc[tk], 0, 0, 0);
     }
 
     if (s + 1 < S) prefetch(s + 1);  // operand regs dead from here
 
-    float dbp = 0.f;
-#pragma unroll
-    for (int i = lane; i < B_; i += 64) dbp += dy_s[i];
-    dbp = wave_sum(dbp);
-    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
+    if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
       if (q == 0 && k < K_)
         ws[k] = round_store<__hip_bfloat16>(ws[k] - lr * gacc[tk][0]);
+      else if (q == 0 && k == K_)
+        ws[32] = round_store<__hip_bfloat16>(bterm - lr * gacc[tk][0]);
     }
-    if (lane == 0) ws[32] = round_store<__hip_bfloat16>(bterm - lr * dbp);
     __syncthreads();  // ws update visible before next iteration's forward
   }
 
@@ -1363,25 +1383,26 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     f32x4 gacc[KT];
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) gacc[tk] = {0.f, 0.f, 0.f, 0.f};
+    // db rides output column K_ of the bwd tile against a constant-1
+    // B-operand (K_ % 16 != 0 guarantees the free column) — same MFMA
+    // chain order as the single-step kernel, so history stays bitwise.
+    static_assert(K_ % 16 != 0, "free db column requires K_ % 16 != 0");
 #pragma unroll
     for (int ii = 0; ii < BS; ++ii)
 #pragma unroll
-      for (int tk = 0; tk < KT; ++tk)
-        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], cbB[tk][ii],
+      for (int tk = 0; tk < KT; ++tk) {
+        const int k = tk * 16 + r;
+        const float bop = (k == K_) ? 1.f : cbB[tk][ii];
+        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x4f32(av[ii], bop,
                                                         gacc[tk], 0, 0, 0);
+      }
 
     // cur regs are dead from here: load next step's operands straight into
     // them — the vmcnt wait attaches to their first use (next iteration's
     // forward MFMA), shadowed by the update/barrier below
     if (s + 1 < S) prefetch(s + 1);
 
-    // db: same lane-strided dy_s reduction order as the single-step
-    // kernel (keeps the multi-step history bitwise-identical)
-    float dbp = 0.f;
-#pragma unroll
-    for (int i = lane; i < B_; i += 64) dbp += dy_s[i];
-    dbp = wave_sum(dbp);
-    if (use_mse && s == S - 1) loss_last = wave_sum(loss_acc);
+    if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
     // no barrier needed here: the dy_s barrier above already ordered this
     // step's ws READS (forward) before these writes
@@ -1389,8 +1410,9 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
       if (q == 0 && k < K_) ws[k] = round_store<T>(ws[k] - lr * gacc[tk][0]);
+      else if (q == 0 && k == K_)
+        ws[32] = round_store<T>(bterm - lr * gacc[tk][0]);
     }
-    if (lane == 0) ws[32] = round_store<T>(bterm - lr * dbp);
     __syncthreads();  // ws update visible before next iteration's forward
   }
 
