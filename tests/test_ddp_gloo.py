@@ -43,6 +43,13 @@ def _reference_params(seed, data):
     return [p.detach().clone() for p in model.parameters()]
 
 
+def _free_port() -> int:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _ddp_worker(rank, seed, data, port, grad_views, bucket_cap_mb, out_path):
     os.environ["MASTER_ADDR"] = "127.0.0.1"
     os.environ["MASTER_PORT"] = str(port)
@@ -95,7 +102,7 @@ def test_ddp_matches_single_process(grad_views):
     seed = 1234
     data = _make_data(seed)
     ref = _reference_params(seed, data)
-    got = _run_ddp(seed, data, port=29611 + int(grad_views), grad_views=grad_views)
+    got = _run_ddp(seed, data, port=_free_port(), grad_views=grad_views)
     for a, b in zip(got, ref):
         assert torch.allclose(a, b, atol=1e-6), (a - b).abs().max()
 
@@ -105,7 +112,7 @@ def test_ddp_many_buckets_ordering():
     seed = 77
     data = _make_data(seed)
     ref = _reference_params(seed, data)
-    got = _run_ddp(seed, data, port=29631, grad_views=True,
+    got = _run_ddp(seed, data, port=_free_port(), grad_views=True,
                    bucket_cap_mb=1e-6)
     for a, b in zip(got, ref):
         assert torch.allclose(a, b, atol=1e-6)

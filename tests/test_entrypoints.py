@@ -10,6 +10,13 @@ import sys
 
 import pytest
 
+def _free_port() -> str:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return str(s.getsockname()[1])
+
+
 ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 TORCHRUN = [sys.executable, "-m", "torch.distributed.run"]
 
@@ -35,7 +42,7 @@ def test_single_gpu_script(tmp_path):
 def test_multigpu_spawn_script(tmp_path):
     r = _run([sys.executable, os.path.join(ROOT, "multigpu.py"), "1", "1"],
              cwd=tmp_path, env_extra={"MI355X_WORLD": "2",
-                                      "MASTER_PORT": "29641"})
+                                      "MASTER_PORT": _free_port()})
     assert r.returncode == 0, r.stderr[-2000:]
     # both ranks print the banner; global ranks 0 and 1
     assert "[GPU0] Epoch 0" in r.stdout and "[GPU1] Epoch 0" in r.stdout
@@ -85,7 +92,7 @@ def test_multinode_torchrun_two_groups_one_host(tmp_path):
     # BASELINE.json config 5 shape: 2 x 2-rank groups on one host via c10d
     # rendezvous (multinode path without real nodes); MSE loss; snapshot.
     script = os.path.join(ROOT, "multinode_torchrun.py")
-    port = "29655"
+    port = _free_port()
     procs = []
     env = dict(os.environ)
     env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
@@ -115,7 +122,7 @@ def test_multinode_torchrun_two_groups_one_host(tmp_path):
 def test_profile_script_tiny(tmp_path):
     r = _run([sys.executable, os.path.join(ROOT, "multigpu_profile.py"), "3"],
              cwd=tmp_path,
-             env_extra={"MI355X_WORLD": "2", "MASTER_PORT": "29643",
+             env_extra={"MI355X_WORLD": "2", "MASTER_PORT": _free_port(),
                         "MI355X_PROFILE_MODEL": "tiny",
                         "MI355X_PROFILE_DATASET": "64",
                         "MI355X_PROFILE_BATCH": "8",
