@@ -22,6 +22,7 @@ from __future__ import annotations
 import argparse
 import json
 import os
+import sys
 import time
 
 import torch
@@ -184,12 +185,25 @@ def main():
         torch.cuda.set_device(device)
 
     comm = None
+    comm_kind = "none"
     dist = torch.distributed
     if world > 1:
         # gloo for rendezvous/barriers only; gradient bytes ride RcclComm
         dist.init_process_group("gloo", rank=rank, world_size=world)
         from mi355x_ddp.parallel.comm import GlooComm, RcclCommAdapter
-        comm = RcclCommAdapter(device) if use_cuda else GlooComm()
+        if use_cuda:
+            try:
+                comm = RcclCommAdapter(device)
+                comm_kind = "rccl"
+            except Exception as e:  # keep the scale run alive, but say so
+                print(f"[bench] RcclComm init FAILED ({e!r}); "
+                      "falling back to gloo transport — numbers are NOT "
+                      "the native RCCL path", file=sys.stderr, flush=True)
+                comm = GlooComm()
+                comm_kind = "gloo-fallback"
+        else:
+            comm = GlooComm()
+            comm_kind = "gloo-cpu"
 
     def barrier():
         if world > 1:
@@ -287,6 +301,7 @@ def main():
                 "in_features": 20,
                 "dataset_size": args.dataset,
                 "parallelism": f"dp{world}",
+                "comm": comm_kind,
                 "engine": ("autograd-cpu" if not use_cuda else
                            "fused" if args.engine == "persistent" and world > 1
                            else args.engine),
