@@ -10,11 +10,13 @@ Multi-GPU (driver runs): python -m torch.distributed.run --nnodes=1 \
     --nproc-per-node N --master-addr 127.0.0.1 --master-port P \
     bench.py --gpus N --steps K --warmup W
 
-Engines (--engine): fused (default; 2 kernels + collective per step),
-graph (fused step captured in a hipGraph), autograd (the generic
-Trainer/DDP path with per-param hooks — the path non-toy models take).
-Synthetic data (random, reference shapes), random-init weights, fp32 (the
-reference's precision).
+Engines (--engine): persistent (default; S steps per kernel launch at
+world 1, falls back to fused when a communicator exists), fused (one
+fused kernel + collective + fused SGD per step), graph (fused step
+captured in a hipGraph), autograd (the generic Trainer/DDP path with
+per-param hooks — the path arbitrary models take).
+Synthetic data (random, reference shapes), random-init weights; fp32 (the
+reference's precision) or --dtype bf16 (BASELINE config 2).
 """
 
 from __future__ import annotations
@@ -177,7 +179,6 @@ def main():
     rank = int(os.environ.get("RANK", 0))
     world = int(os.environ.get("WORLD_SIZE", 1))
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
-    n_gpus = max(world, args.gpus)
 
     use_cuda = torch.cuda.is_available()
     device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")

@@ -6,8 +6,13 @@ MI355X_FIRST_BUCKET_MB  first-bucket cap so the first all-reduce fires
                         early (default 1)
 MI355X_GRAD_VIEWS       1 = grads are views into flat buckets (zero-copy,
                         default); 0 = flatten-kernel gather per bucket
-MI355X_HIPGRAPH         1 = capture the steady-state train step in a
-                        hipGraph and replay it (bench fast path)
+MI355X_SHUFFLE          'randperm' = torch.randperm epoch shuffle in
+                        bench.py's device data path (default: fused
+                        in-kernel bijection, ops.perm_index)
+MI355X_EPOCH_SIDE_STREAM  1 = gather epoch shards on a side stream
+                        (measured slower; default 0)
+MI355X_PREFETCH         0 = disable the Trainer's copy-stream H2D
+                        prefetcher (default on for GPU)
 """
 
 import os
