@@ -191,7 +191,8 @@ def main():
     if world > 1:
         # gloo for rendezvous/barriers only; gradient bytes ride RcclComm
         dist.init_process_group("gloo", rank=rank, world_size=world)
-        from mi355x_ddp.parallel.comm import GlooComm, RcclCommAdapter
+        from mi355x_ddp.parallel.comm import (GlooComm, P2pMeshComm,
+                                              RcclCommAdapter)
         if use_cuda:
             try:
                 comm = RcclCommAdapter(device)
@@ -202,6 +203,16 @@ def main():
                       "the native RCCL path", file=sys.stderr, flush=True)
                 comm = GlooComm()
                 comm_kind = "gloo-fallback"
+            if os.environ.get("MI355X_P2P", "1") != "0":
+                # device-side xGMI mesh for the 84 B all-reduce; validated
+                # against gloo at setup, raises (never hangs) on timeout
+                try:
+                    comm = P2pMeshComm(device, base=comm)
+                    comm_kind = "p2p-mesh+" + comm_kind
+                except Exception as e:
+                    print(f"[bench] P2pMesh unavailable ({e!r}); "
+                          f"staying on {comm_kind}", file=sys.stderr,
+                          flush=True)
         else:
             comm = GlooComm()
             comm_kind = "gloo-cpu"
@@ -241,10 +252,15 @@ def main():
                 data.bound_epoch = cur = e
             step_shard(i)
 
+    def comm_check():
+        if comm is not None and hasattr(comm, "check"):
+            comm.check()  # raises if a mesh all-reduce ever timed out
+
     # -- warmup (untimed) -------------------------------------------------
     run_steps(0, args.warmup)
     flush_fn()
     barrier()
+    comm_check()
 
     # -- timed region: exactly K steps (any deferred launches are flushed
     #    INSIDE the bracket — all K steps' work executes before the
@@ -254,6 +270,7 @@ def main():
     flush_fn()
     barrier()
     elapsed = time.perf_counter() - t0
+    comm_check()
 
     # max over ranks (gloo all-reduce of the scalar)
     if world > 1:

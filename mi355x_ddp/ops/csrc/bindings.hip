@@ -2,6 +2,7 @@
 #include <torch/extension.h>
 
 #include "ops.h"
+#include "p2p_mesh.h"
 #include "rccl_comm.h"
 
 namespace py = pybind11;
@@ -42,6 +43,21 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("param_flat"), py::arg("loss_out"), py::arg("use_mse") = true,
         py::arg("w_off") = 0, py::arg("b_off") = 0, py::arg("lr") = 0.0,
         py::arg("batch") = 32);
+
+  py::class_<mi355x::P2pMesh>(m, "P2pMesh")
+      .def(py::init<int, int, int>(), py::arg("rank"), py::arg("world"),
+           py::arg("device"))
+      .def("handle_bytes",
+           [](const mi355x::P2pMesh& m_) { return py::bytes(m_.handle_bytes()); })
+      .def("connect",
+           [](mi355x::P2pMesh& m_, const std::vector<py::bytes>& hs) {
+             std::vector<std::string> v(hs.begin(), hs.end());
+             m_.connect(v);
+           })
+      .def("all_reduce_avg_inline", &mi355x::P2pMesh::all_reduce_avg_inline)
+      .def("check", &mi355x::P2pMesh::check)
+      .def_property_readonly("rank", &mi355x::P2pMesh::rank)
+      .def_property_readonly("world", &mi355x::P2pMesh::world);
 
   py::class_<mi355x::RcclComm>(m, "RcclComm")
       .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),

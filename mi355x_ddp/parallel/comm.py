@@ -114,6 +114,76 @@ class RcclCommAdapter:
         self._comm.barrier()
 
 
+class P2pMeshComm:
+    """Device-side xGMI mesh all-reduce for TINY payloads (the toy step's
+    84 B of gradients), layered over a base communicator that keeps
+    handling broadcast / barrier / large tensors.
+
+    Setup exchanges HIP IPC mailbox handles through the c10d store, then
+    `validate()` cross-checks one mesh all-reduce against the gloo
+    process group; any mismatch or kernel timeout raises, so callers can
+    fall back to the base (RCCL) transport. The mesh kernel's spin-wait is
+    wall-clock-bounded — a lost peer surfaces as a raised error, never a
+    hang."""
+
+    _KEY = "mi355x_ddp/p2p_mesh"
+
+    def __init__(self, device: torch.device, base):
+        from .. import ops
+        assert dist.is_initialized()
+        self.base = base
+        self.rank = dist.get_rank()
+        self.world = dist.get_world_size()
+        self._mesh = ops.ext().P2pMesh(self.rank, self.world,
+                                       device.index or 0)
+        store = dist.distributed_c10d._get_default_store()
+        store.set(f"{self._KEY}/{self.rank}",
+                  base64.b64encode(self._mesh.handle_bytes()).decode())
+        handles = []
+        for r in range(self.world):
+            handles.append(base64.b64decode(store.get(f"{self._KEY}/{r}")))
+        self._mesh.connect(handles)
+        self._device = device
+        self.validate()
+
+    def validate(self) -> None:
+        """One mesh all-reduce cross-checked against the gloo group."""
+        probe = torch.arange(24, dtype=torch.float32, device=self._device)
+        probe = probe * (self.rank + 1)
+        ref = probe.cpu()
+        dist.all_reduce(ref, op=dist.ReduceOp.SUM)
+        ref = ref / self.world
+        self._mesh.all_reduce_avg_inline(probe)
+        torch.cuda.synchronize()
+        self._mesh.check()
+        if not torch.allclose(probe.cpu(), ref, atol=1e-5):
+            raise RuntimeError(
+                "P2pMesh validation mismatch vs gloo all-reduce: "
+                f"{(probe.cpu() - ref).abs().max().item()}")
+
+    def check(self) -> None:
+        self._mesh.check()
+
+    def all_reduce_avg_inline(self, t: torch.Tensor) -> None:
+        if t.numel() <= 64:
+            self._mesh.all_reduce_avg_inline(t)
+        else:
+            self.base.all_reduce_avg_inline(t)
+
+    # everything else rides the base transport
+    def all_reduce_avg(self, t: torch.Tensor) -> None:
+        self.base.all_reduce_avg(t)
+
+    def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
+        self.base.broadcast(t, root)
+
+    def join_compute(self) -> None:
+        self.base.join_compute()
+
+    def barrier(self) -> None:
+        self.base.barrier()
+
+
 def create_comm(device: torch.device):
     """Pick the communicator for this process, or None when world size is 1
     (single-process runs use the same reducer with no collectives)."""

@@ -18,6 +18,7 @@ ext = CUDAExtension(
     sources=[
         "mi355x_ddp/ops/csrc/kernels.hip",
         "mi355x_ddp/ops/csrc/rccl_comm.hip",
+        "mi355x_ddp/ops/csrc/p2p_mesh.hip",
         "mi355x_ddp/ops/csrc/bindings.hip",
     ],
     libraries=["rccl"],
