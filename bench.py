@@ -164,8 +164,9 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
             opt.step()
         return step, noflush, None
 
-    if kind == "persistent" and comm is not None:
-        kind = "fused"  # multi-step kernel is the world-1 in-kernel-SGD path
+    if (kind == "persistent" and comm is not None
+            and getattr(comm, "_mesh", None) is None):
+        kind = "fused"  # without a mesh comm, multi-step is world-1 only
     cls = {"persistent": PersistentToyStep, "graph": GraphedToyStep,
            "fused": ToyFusedStep}[kind]
     eng = cls(model, comm=comm, lr=lr, use_mse=True)
@@ -181,6 +182,8 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", 0))
 
     use_cuda = torch.cuda.is_available()
+    if os.environ.get("MI355X_FORCE_DEV0") == "1":
+        local_rank = 0  # test-only: run a multi-rank world on ONE device
     device = torch.device("cuda", local_rank) if use_cuda else torch.device("cpu")
     if use_cuda:
         torch.cuda.set_device(device)
@@ -321,8 +324,11 @@ def main():
                 "parallelism": f"dp{world}",
                 "comm": comm_kind,
                 "engine": ("autograd-cpu" if not use_cuda else
-                           "fused" if args.engine == "persistent" and world > 1
-                           else args.engine),
+                           "autograd" if engine_obj is None else
+                           {"PersistentToyStep": "persistent",
+                            "GraphedToyStep": "graph",
+                            "ToyFusedStep": "fused"}[
+                               type(engine_obj).__name__]),
                 "loss": "mse",
                 "p50_step_ms": p50_ms,
             },

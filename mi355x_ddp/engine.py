@@ -87,8 +87,11 @@ class PersistentToyStep(ToyFusedStep):
 
     def __init__(self, *args, max_defer: int = 1024, **kwargs):
         super().__init__(*args, **kwargs)
-        assert self.comm is None, \
-            "PersistentToyStep is the world-1 path (in-kernel SGD)"
+        # world-1 (no comm), or a mesh-capable comm: the in-kernel xGMI
+        # mesh exchange keeps SGD in-kernel at any world size
+        self._mesh = getattr(self.comm, "_mesh", None)
+        assert self.comm is None or self._mesh is not None, \
+            "PersistentToyStep needs world 1 or a P2pMeshComm"
         self.max_defer = max_defer
         self._install_fast_path()
 
@@ -102,7 +105,25 @@ class PersistentToyStep(ToyFusedStep):
         loss_out = self.loss_out
         use_mse, w_off, b_off = self.use_mse, self.w_off, self.b_off
         lr, max_defer = self.lr, self.max_defer
-        eager = ToyFusedStep.step.__get__(self)
+        mesh = self._mesh
+
+        def launch(xall, tall, B):
+            """One deferred-run launch: world-1 multistep, or the mesh
+            variant with an in-kernel all-reduce per step."""
+            if mesh is None:
+                ext_mod.toy_multistep(xall, tall, flat_param, loss_out,
+                                      use_mse, w_off, b_off, lr, B)
+            else:
+                ext_mod.toy_multistep_mesh(xall, tall, flat_param, loss_out,
+                                           use_mse, w_off, b_off, lr, B,
+                                           mesh)
+
+        if mesh is None:
+            eager = ToyFusedStep.step.__get__(self)
+        else:
+            def eager(x, t):  # single-step via the same mesh kernel (S=1):
+                # keeps the per-step collective order identical on all ranks
+                launch(x.contiguous(), t.contiguous(), x.shape[0])
         x0 = t0 = None
         shape = None
         count = nx = nt = bk = bt = 0
@@ -170,15 +191,11 @@ class PersistentToyStep(ToyFusedStep):
                     xall = xx.as_strided((n * B, K), (K, 1))
                     tall = tt.as_strided((n * B,) + tt.shape[1:],
                                          (tt.stride(0),) + tt.stride()[1:])
-                    ext_mod.toy_multistep(xall, tall, flat_param, loss_out,
-                                          use_mse, w_off, b_off, lr, B)
+                    launch(xall, tall, B)
             if s_next > s_start:
                 lo, n = s_start * sbatch, (s_next - s_start) * sbatch
                 s_start = s_next
-                xx = shard_x[lo:lo + n]
-                tt = shard_t[lo:lo + n]
-                ext_mod.toy_multistep(xx, tt, flat_param, loss_out,
-                                      use_mse, w_off, b_off, lr, sbatch)
+                launch(shard_x[lo:lo + n], shard_t[lo:lo + n], sbatch)
 
         self.step = step
         self.flush = flush

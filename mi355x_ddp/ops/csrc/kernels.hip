@@ -24,6 +24,7 @@
 #include <c10/hip/HIPStream.h>
 
 #include "ops.h"
+#include "p2p_mesh.h"
 
 #define HIP_OK(expr)                                                          \
   do {                                                                        \
@@ -1116,6 +1117,63 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
 // by r==0 lanes) and the per-row dY exchange between the forward readout
 // and the backward MFMA. Divisions are hoisted to one reciprocal
 // (2/B is exact for the reference batch 32, so dy is bitwise-unchanged).
+
+// In-kernel mesh exchange for the multi-step trainers: scatter this
+// rank's 21-float gradient vector (held one value per q==0 lane per
+// K-tile, db on the free column) into every rank's mailbox, publish a
+// per-step sequence number, bounded-wait for all peers, and return the
+// averaged value for this lane's tile. Returns false on timeout (caller
+// must set the error flag and exit).
+template <int KT>
+__device__ __forceinline__ bool mesh_exchange(
+    float (&gval)[KT], int K_, int lane, int r, int q,
+    MeshSlot* const* __restrict__ peer_slots, MeshSlot* __restrict__ my_mb,
+    int mworld, float minv_world, unsigned long long sq) {
+  if (q == 0) {
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      if (k <= K_) {
+        for (int p = 0; p < mworld; ++p) peer_slots[p]->data[k] = gval[tk];
+      }
+    }
+  }
+  __threadfence_system();
+  __syncthreads();
+  if (lane == 0) {
+    for (int p = 0; p < mworld; ++p)
+      __hip_atomic_store(&peer_slots[p]->seq, sq, __ATOMIC_RELEASE,
+                         __HIP_MEMORY_SCOPE_SYSTEM);
+  }
+  bool timed_out = false;
+  if (lane < mworld) {
+    const unsigned long long t0c = __builtin_amdgcn_s_memrealtime();
+    while (__hip_atomic_load(&my_mb[lane].seq, __ATOMIC_ACQUIRE,
+                             __HIP_MEMORY_SCOPE_SYSTEM) < sq) {
+      if (__builtin_amdgcn_s_memrealtime() - t0c > 25000000ull) {
+        timed_out = true;
+        break;
+      }
+      __builtin_amdgcn_s_sleep(8);
+    }
+  }
+  if (__any(timed_out)) return false;
+  __threadfence();
+  __syncthreads();
+  if (q == 0) {
+#pragma unroll
+    for (int tk = 0; tk < KT; ++tk) {
+      const int k = tk * 16 + r;
+      if (k <= K_) {
+        float sum = 0.f;
+        for (int p = 0; p < mworld; ++p) sum += my_mb[p].data[k];
+        gval[tk] = sum * minv_world;
+      }
+    }
+  }
+  return true;
+}
+
 // ---------------------------------------------------------------------------
 // bf16 wide-MFMA variant: the whole K (<=32) contraction of the forward and
 // the whole B (=32) contraction of the backward are each ONE
@@ -1127,13 +1185,18 @@ __global__ void k_toy_multistep(const T* __restrict__ X,
 // Operand map (as k_gemm_bf16): A[l&15][(l>>4)*8+j]; B[(l>>4)*8+j][l&15];
 // D col=l&15, row=(l>>4)*4+reg.
 // ---------------------------------------------------------------------------
-template <int B_, int K_>
+template <int B_, int K_, bool MESH = false>
 __global__ void __launch_bounds__(64, 1)
 k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
                       const __hip_bfloat16* __restrict__ Tg,
                       __hip_bfloat16* __restrict__ param,
                       float* __restrict__ loss_out,
-                      int S, int use_mse, int w_off, int b_off, float lr) {
+                      int S, int use_mse, int w_off, int b_off, float lr,
+                      MeshSlot* const* __restrict__ peer_slots = nullptr,
+                      MeshSlot* __restrict__ my_mb = nullptr,
+                      int mworld = 1, float minv_world = 1.f,
+                      unsigned long long seq0 = 0,
+                      unsigned int* __restrict__ mesh_err = nullptr) {
   static_assert(B_ == 32 && K_ <= 32, "bf16 wide path: B=32, K<=32");
   constexpr int MT = 2;                // fwd 16-row tiles
   constexpr int KT = (K_ + 15) / 16;   // bwd 16-col tiles
@@ -1246,6 +1309,20 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
 
     if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
+    if constexpr (MESH) {
+      float gval[KT];
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) gval[tk] = gacc[tk][0];
+      if (!mesh_exchange<KT>(gval, K_, lane, r, q, peer_slots, my_mb,
+                             mworld, minv_world,
+                             seq0 + (unsigned long long)s)) {
+        if (lane == 0) *mesh_err = 1u;
+        return;
+      }
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) gacc[tk][0] = gval[tk];
+    }
+
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
@@ -1262,11 +1339,16 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
   if (lane == 0 && loss_out) *loss_out = use_mse ? loss_last / (float)B_ : 0.f;
 }
 
-template <typename T, int B_, int K_>
+template <typename T, int B_, int K_, bool MESH = false>
 __global__ void __launch_bounds__(64, 1)
 k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
                      T* __restrict__ param, float* __restrict__ loss_out,
-                     int S, int use_mse, int w_off, int b_off, float lr) {
+                     int S, int use_mse, int w_off, int b_off, float lr,
+                     MeshSlot* const* __restrict__ peer_slots = nullptr,
+                     MeshSlot* __restrict__ my_mb = nullptr,
+                     int mworld = 1, float minv_world = 1.f,
+                     unsigned long long seq0 = 0,
+                     unsigned int* __restrict__ mesh_err = nullptr) {
   constexpr int MT = (B_ + 15) / 16;   // fwd 16-row tiles
   constexpr int KT = (K_ + 15) / 16;   // bwd 16-col tiles
   constexpr int KS = (K_ + 3) / 4;     // fwd k-steps
@@ -1404,6 +1486,20 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
 
     if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
+    if constexpr (MESH) {
+      float gval[KT];
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) gval[tk] = gacc[tk][0];
+      if (!mesh_exchange<KT>(gval, K_, lane, r, q, peer_slots, my_mb,
+                             mworld, minv_world,
+                             seq0 + (unsigned long long)s)) {
+        if (lane == 0) *mesh_err = 1u;
+        return;
+      }
+#pragma unroll
+      for (int tk = 0; tk < KT; ++tk) gacc[tk][0] = gval[tk];
+    }
+
     // no barrier needed here: the dy_s barrier above already ordered this
     // step's ws READS (forward) before these writes
 #pragma unroll
@@ -1485,6 +1581,46 @@ void toy_multistep(torch::Tensor x, torch::Tensor t, torch::Tensor param_flat,
     launch_toy_multistep<scalar_t>(x, t, param_flat, lossp, use_mse,
                                    (int)w_off, (int)b_off, (float)lr, B, K, S);
   });
+  HIP_OK(hipGetLastError());
+}
+
+void toy_multistep_mesh(torch::Tensor x, torch::Tensor t,
+                        torch::Tensor param_flat, torch::Tensor loss_out,
+                        bool use_mse, int64_t w_off, int64_t b_off, double lr,
+                        int64_t batch, P2pMesh& mesh) {
+  const int B = (int)batch, K = (int)x.size(1);
+  TORCH_CHECK(B == 32 && K == 20,
+              "mesh multistep supports the reference shape (batch 32, K 20)");
+  TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
+  TORCH_CHECK(x.size(0) % B == 0 && t.size(0) == x.size(0));
+  TORCH_CHECK(lr > 0.0, "mesh multistep applies SGD in-kernel");
+  const int S = (int)(x.size(0) / B);
+  if (S == 0) return;
+  float* lossp = nullptr;
+  if (loss_out.defined() && loss_out.numel()) {
+    TORCH_CHECK(loss_out.scalar_type() == at::kFloat, "loss_out must be f32");
+    lossp = loss_out.data_ptr<float>();
+  }
+  const unsigned long long seq0 = mesh.alloc_seq((unsigned long long)S);
+  const float inv = 1.f / (float)mesh.world();
+  if (x.scalar_type() == at::kFloat) {
+    hipLaunchKernelGGL((k_toy_multistep_spec<float, 32, 20, true>), dim3(1),
+                       dim3(64), 0, cur_stream(), cdptr<float>(x),
+                       cdptr<float>(t), dptr<float>(param_flat), lossp, S,
+                       use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
+                       mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
+                       seq0, mesh.err_flag());
+  } else if (x.scalar_type() == at::kBFloat16) {
+    hipLaunchKernelGGL((k_toy_multistep_bf16w<32, 20, true>), dim3(1),
+                       dim3(64), 0, cur_stream(),
+                       cdptr<__hip_bfloat16>(x), cdptr<__hip_bfloat16>(t),
+                       dptr<__hip_bfloat16>(param_flat), lossp, S,
+                       use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
+                       mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
+                       seq0, mesh.err_flag());
+  } else {
+    TORCH_CHECK(false, "mesh multistep: dtype must be f32 or bf16");
+  }
   HIP_OK(hipGetLastError());
 }
 

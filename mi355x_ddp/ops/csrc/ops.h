@@ -4,6 +4,8 @@
 #include <torch/extension.h>
 #include <cstdint>
 
+#include "p2p_mesh.h"
+
 namespace mi355x {
 
 // y = x @ w^T + b          (reference call site: model(source), single_gpu.py:23)
@@ -76,5 +78,13 @@ std::vector<torch::Tensor> epoch_shard(torch::Tensor X, torch::Tensor Tg,
 void toy_multistep(torch::Tensor x, torch::Tensor t, torch::Tensor param_flat,
                    torch::Tensor loss_out, bool use_mse,
                    int64_t w_off, int64_t b_off, double lr, int64_t batch);
+
+// Multi-step trainer with an IN-KERNEL xGMI mesh all-reduce per step
+// (world > 1): the deferred-launch engine at any world size, one
+// collective exchange per step without leaving the kernel.
+void toy_multistep_mesh(torch::Tensor x, torch::Tensor t,
+                        torch::Tensor param_flat, torch::Tensor loss_out,
+                        bool use_mse, int64_t w_off, int64_t b_off, double lr,
+                        int64_t batch, P2pMesh& mesh);
 
 }  // namespace mi355x

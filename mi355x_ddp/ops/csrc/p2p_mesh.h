@@ -52,6 +52,19 @@ class P2pMesh {
   int rank() const { return rank_; }
   int world() const { return world_; }
 
+  // device-view accessors for kernels that embed the mesh exchange
+  // (toy_multistep_mesh in kernels.hip)
+  MeshSlot* my_mb() const { return my_mb_; }
+  MeshSlot* const* peer_slots() const { return peer_slot_dev_; }
+  unsigned int* err_flag() const { return err_host_; }
+  // reserve n sequence numbers; returns the first. Every rank must make
+  // identical reservation sequences (they do: same engine code path).
+  unsigned long long alloc_seq(unsigned long long n) {
+    unsigned long long first = seq_ + 1;
+    seq_ += n;
+    return first;
+  }
+
  private:
   int rank_, world_, device_;
   MeshSlot* my_mb_ = nullptr;          // device, fine-grained, world slots

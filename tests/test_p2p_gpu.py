@@ -107,3 +107,47 @@ def test_fused_engine_world2_mesh_matches_gloo(tmp_path):
     assert torch.allclose(a["w"], b["w"], atol=1e-6), \
         (a["w"] - b["w"]).abs().max()
     assert torch.allclose(a["b"], b["b"], atol=1e-6)
+
+
+def _persistent_worker(rank, world, port, out_dir):
+    from mi355x_ddp.engine import PersistentToyStep
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel.comm import GlooComm, P2pMeshComm
+    _init(rank, world, port)
+    try:
+        comm = P2pMeshComm(torch.device("cuda", 0), base=GlooComm())
+        torch.manual_seed(11)
+        model = toy_model(20, 1).to("cuda")
+        eng = PersistentToyStep(model, comm=comm, lr=0.05, use_mse=True)
+        eng.reducer.broadcast_params(root=0)
+        g = torch.Generator().manual_seed(100 + rank)
+        Xf = torch.rand(12 * 32, 20, generator=g).to("cuda")
+        Tf = torch.rand(12 * 32, 1, generator=g).to("cuda")
+        eng.bind_shard(Xf, Tf, 32)
+        for s in range(12):
+            eng.step_shard(s)
+        eng.flush()
+        torch.cuda.synchronize()
+        comm.check()
+        if rank == 0:
+            torch.save({"w": model.weight.detach().cpu(),
+                        "b": model.bias.detach().cpu()},
+                       os.path.join(out_dir, "persistent_mesh.pt"))
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_persistent_mesh_world2_matches_fused_gloo(tmp_path):
+    # same data/ordering as _train_worker's gloo run: deferred multi-step
+    # kernel with an IN-KERNEL mesh all-reduce per step == per-step fused
+    # kernel + gloo all-reduce
+    mp.spawn(_train_worker, args=(2, _free_port(), False, str(tmp_path)),
+             nprocs=2, join=True)
+    mp.spawn(_persistent_worker, args=(2, _free_port(), str(tmp_path)),
+             nprocs=2, join=True)
+    a = torch.load(tmp_path / "mesh0.pt", weights_only=True)
+    b = torch.load(tmp_path / "persistent_mesh.pt", weights_only=True)
+    assert torch.allclose(a["w"], b["w"], atol=1e-5), \
+        (a["w"] - b["w"]).abs().max()
+    assert torch.allclose(a["b"], b["b"], atol=1e-5)
