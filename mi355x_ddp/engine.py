@@ -70,8 +70,11 @@ class ToyFusedStep:
 
 
 class PersistentToyStep(ToyFusedStep):
-    """World-1 engine: runs of consecutive steps execute as ONE multi-step
-    kernel launch (`toy_multistep`), weights resident in LDS across steps.
+    """Deferred-launch engine: runs of consecutive steps execute as ONE
+    multi-step kernel launch (`toy_multistep`), weights resident in LDS
+    across steps. Works at world 1 (no communicator) and, with a
+    `P2pMeshComm`, at any world size — the kernel then performs one
+    in-kernel xGMI mesh all-reduce per step (`toy_multistep_mesh`).
 
     `step(x, t)` defers when the incoming batch is the next contiguous
     [B, K] slice of the same device buffer (the device-resident epoch

@@ -151,3 +151,39 @@ def test_persistent_mesh_world2_matches_fused_gloo(tmp_path):
     assert torch.allclose(a["w"], b["w"], atol=1e-5), \
         (a["w"] - b["w"]).abs().max()
     assert torch.allclose(a["b"], b["b"], atol=1e-5)
+
+
+def _persistent_bf16_worker(rank, world, port, out_dir):
+    from mi355x_ddp.engine import PersistentToyStep
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel.comm import GlooComm, P2pMeshComm
+    _init(rank, world, port)
+    try:
+        comm = P2pMeshComm(torch.device("cuda", 0), base=GlooComm())
+        torch.manual_seed(11)
+        model = toy_model(20, 1).to("cuda").bfloat16()
+        eng = PersistentToyStep(model, comm=comm, lr=0.05, use_mse=True)
+        eng.reducer.broadcast_params(root=0)
+        g = torch.Generator().manual_seed(100 + rank)
+        Xf = torch.rand(10 * 32, 20, generator=g).to("cuda").bfloat16()
+        Tf = torch.rand(10 * 32, 1, generator=g).to("cuda").bfloat16()
+        eng.bind_shard(Xf, Tf, 32)
+        for s in range(10):
+            eng.step_shard(s)
+        eng.flush()
+        torch.cuda.synchronize()
+        comm.check()
+        w = model.weight.detach().float().cpu()
+        assert torch.isfinite(w).all()
+        if rank == 0:
+            torch.save({"w": w}, os.path.join(out_dir, "bf16.pt"))
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_persistent_mesh_bf16_world2(tmp_path):
+    mp.spawn(_persistent_bf16_worker, args=(2, _free_port(), str(tmp_path)),
+             nprocs=2, join=True)
+    w = torch.load(tmp_path / "bf16.pt", weights_only=True)["w"]
+    assert torch.isfinite(w).all() and w.abs().sum() > 0
