@@ -1435,15 +1435,8 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
         acc[tm] = __builtin_amdgcn_mfma_f32_16x16x4f32(cfA[tm][kk], wv[kk],
                                                        acc[tm], 0, 0, 0);
 
-    // ---- loss grad + dY exchange, register-only ----
-    // dY producers AND consumers are the r==0 lanes (0,16,32,48): row
-    // tm*16+q*4+i lives on lane 16*q, and the backward A-operand needs
-    // dy[4*ii+q] — i.e. register i=q of lane 16*(ii&3), tile ii>=4. So
-    // the exchange is four constant-lane broadcasts per tile, no LDS
-    // round trip and no barrier (same float values: bitwise-unchanged).
-    static_assert(B_ == 32, "register dY exchange is laid out for B=32");
+    // ---- loss grad (rows live on r==0 lanes) + dY exchange ----
     float loss_acc = 0.f;
-    float myDy[MT][4];
     if (r == 0) {
 #pragma unroll
       for (int tm = 0; tm < MT; ++tm)
@@ -1456,24 +1449,18 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
             loss_acc += d * d;
             dy = d * inv2B;
           }
-          myDy[tm][i] = dy;
+          if (row < B_) dy_s[row] = dy;
         }
     }
-    // select this lane's q-th register of each tile, then broadcast from
-    // the four producer lanes
-    float dqa = myDy[0][0], dqb = myDy[1][0];
-#pragma unroll
-    for (int j = 1; j < 4; ++j) {
-      dqa = (q == j) ? myDy[0][j] : dqa;
-      dqb = (q == j) ? myDy[1][j] : dqb;
-    }
+    __syncthreads();  // dy_s visible to all lanes
 
     // ---- backward: dw_k = sum_i dY_i X[i,k] ----
     float av[BS];
 #pragma unroll
     for (int ii = 0; ii < BS; ++ii) {
-      const float v = __shfl((ii >= 4) ? dqb : dqa, 16 * (ii & 3), 64);
-      av[ii] = (r == 0) ? v : 0.f;
+      const int i = 4 * ii + q;
+      const float v = dy_s[(i < B_) ? i : B_ - 1];
+      av[ii] = (r == 0 && i < B_) ? v : 0.f;
     }
     f32x4 gacc[KT];
 #pragma unroll
