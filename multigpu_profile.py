@@ -38,9 +38,14 @@ def _tiny_cnn():
 
 def load_train_objs(dataset_size: int = 2048):
     shape = (3, 224, 224)
-    if os.environ.get("MI355X_PROFILE_MODEL") == "tiny":
+    kind = os.environ.get("MI355X_PROFILE_MODEL", "resnet50")
+    if kind == "tiny":
         shape = (3, 32, 32)
         model = _tiny_cnn()
+    elif kind == "vit":
+        from mi355x_ddp.models import vit_l_32
+        model = vit_l_32()  # the reference imports it but leaves it
+        # commented out (ref multigpu_profile.py:24); here it runs
     else:
         model = resnet50()
     train_set = RandomImageDataset(dataset_size, shape)
