@@ -78,7 +78,13 @@ class Trainer:
         profile_dir: str = "./log/resnet50/",
         save_wrapped: bool = False,
         bucket_cap_mb: Optional[float] = None,
+        engine: str = "hooks",
     ) -> None:
+        """engine: "hooks" (default — the generic autograd+reducer path,
+        reference semantics for arbitrary models), or "fused"/"persistent"
+        /"auto" to run the toy fast path (single-launch fused step /
+        multi-step deferred kernel) when the model+loss+optimizer qualify;
+        silently falls back to hooks otherwise."""
         if gpu_id is None:
             gpu_id = int(os.environ.get("LOCAL_RANK", 0))
         self.gpu_id = gpu_id
@@ -115,8 +121,10 @@ class Trainer:
             self._load_snapshot(snapshot_path)
 
         self._distributed = False
-        import torch.distributed as dist
-        if wrap_ddp and dist.is_available() and dist.is_initialized():
+        self._engine = None
+        if engine != "hooks" and self._try_fast_engine(engine):
+            pass  # _run_batch drives self._engine
+        elif wrap_ddp and dist.is_available() and dist.is_initialized():
             # world 1 wraps too: the engine still provides flat buckets and
             # the fused SGD path (there is simply no communicator).
             from .parallel import DDP, FusedSGD
@@ -124,6 +132,36 @@ class Trainer:
             self._distributed = True
             if isinstance(self.optimizer, FusedSGD):
                 self.optimizer.attach_reducer(self.model.reducer)
+
+    def _try_fast_engine(self, kind: str) -> bool:
+        """Engage the toy fast path when the configuration qualifies:
+        HipLinear(K,1) model, MSE loss, FusedSGD with a single lr, CUDA
+        device. Returns False (caller keeps the hooks path) otherwise."""
+        import torch.distributed as dist
+        from .models.toy import HipLinear
+        from .parallel import FusedSGD
+        from .parallel.comm import create_comm
+        m = self.model
+        if not (isinstance(m, HipLinear) and m.out_features == 1
+                and self.device.type == "cuda"
+                and self.loss_fn is ops.mse_loss
+                and isinstance(self.optimizer, FusedSGD)
+                and len(self.optimizer.param_groups) == 1):
+            return False
+        comm = create_comm(self.device)
+        if kind == "auto":
+            kind = "persistent" if comm is None else "fused"
+        from .engine import PersistentToyStep, ToyFusedStep
+        lr = self.optimizer.param_groups[0]["lr"]
+        if kind == "persistent":
+            if comm is not None and getattr(comm, "_mesh", None) is None:
+                kind = "fused"  # multi-step needs world 1 or a mesh comm
+        cls = PersistentToyStep if kind == "persistent" else ToyFusedStep
+        self._engine = cls(m, comm=comm, lr=lr, use_mse=True)
+        if comm is not None:
+            self._engine.reducer.broadcast_params(root=0)
+            self._distributed = True
+        return True
 
     # -- snapshot / checkpoint -------------------------------------------
     def _unwrapped(self) -> torch.nn.Module:
@@ -159,6 +197,9 @@ class Trainer:
         # Reference single_gpu.py:21-26. zero_grad is folded into the
         # bucket lifecycle when the model is DDP-wrapped (the fused SGD
         # zeroes the flat grad buffer); otherwise we zero here.
+        if self._engine is not None:
+            self._engine.step(source.contiguous(), targets.contiguous())
+            return
         if not self._distributed:
             self.optimizer.zero_grad(set_to_none=False)
         output = self.model(source)
@@ -212,6 +253,8 @@ class Trainer:
             self._run_batch(source, targets)
             if self._profiler is not None:
                 self._profiler.step()
+        if self._engine is not None and hasattr(self._engine, "flush"):
+            self._engine.flush()  # deferred engines: run any pending steps
 
     def _create_profiler(self):
         # Parity with reference multigpu_profile.py:80-91: schedule

@@ -207,3 +207,31 @@ def test_shard_bound_non_sequential_indices():
                        ref_model.weight.detach().cpu())
     assert torch.equal(model.bias.detach().cpu(),
                        ref_model.bias.detach().cpu())
+
+
+def test_trainer_fast_engine_matches_hooks_on_gpu(tmp_path):
+    # Trainer(engine="persistent") == Trainer(hooks) on the toy workload
+    from mi355x_ddp.data import ToyDataset, prepare_dataloader
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.trainer import Trainer
+
+    def run(engine):
+        torch.manual_seed(3)
+        model = toy_model(20, 1)
+        loader = prepare_dataloader(ToyDataset(256, seed=5), 32,
+                                    shuffle=False)
+        opt = FusedSGD(model.parameters(), lr=0.05)
+        tr = Trainer(model, loader, opt, 0, save_every=10**9, loss_fn="mse",
+                     wrap_ddp=False, engine=engine,
+                     checkpoint_path=str(tmp_path / f"{engine}.pt"))
+        if engine != "hooks":
+            assert tr._engine is not None
+        tr.train(2)
+        torch.cuda.synchronize()
+        return model.weight.detach().cpu().clone()
+
+    w_hooks = run("hooks")
+    w_fast = run("persistent")
+    assert torch.allclose(w_fast, w_hooks, atol=1e-5), \
+        (w_fast - w_hooks).abs().max()

@@ -91,3 +91,23 @@ def test_ce_loss_multiclass_matches_torch():
     loss = ops.cross_entropy(y, t)
     ref = torch.nn.CrossEntropyLoss()(y.detach(), t)
     assert torch.allclose(loss, ref, atol=1e-6)
+
+
+def test_trainer_fast_engine_falls_back_on_cpu(tmp_path):
+    # engine="auto" on a CPU host: conditions unmet (no CUDA) -> hooks path
+    import torch
+    from mi355x_ddp.data import ToyDataset, prepare_dataloader
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.trainer import Trainer
+    torch.manual_seed(0)
+    model = toy_model(20, 1)
+    loader = prepare_dataloader(ToyDataset(128, seed=1), 32, shuffle=False)
+    opt = FusedSGD(model.parameters(), lr=1e-3)
+    tr = Trainer(model, loader, opt, "cpu", save_every=10**9, loss_fn="mse",
+                 wrap_ddp=False, engine="auto",
+                 checkpoint_path=str(tmp_path / "c.pt"))
+    assert tr._engine is None  # fell back
+    w0 = model.weight.detach().clone()
+    tr.train(1)
+    assert not torch.equal(model.weight.detach(), w0)
