@@ -264,6 +264,9 @@ def main():
     flush_fn()
     barrier()
     comm_check()
+    if comm is not None and hasattr(comm, "validate"):
+        comm.validate()  # mesh transport re-verified against gloo after
+        # warmup — catches coherence problems before anything is timed
 
     # -- timed region: exactly K steps (any deferred launches are flushed
     #    INSIDE the bracket — all K steps' work executes before the
