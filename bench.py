@@ -260,6 +260,7 @@ def main():
             comm.check()  # raises if a mesh all-reduce ever timed out
 
     # -- warmup (untimed) -------------------------------------------------
+    barrier()  # align ranks before the first mesh-synchronized step
     run_steps(0, args.warmup)
     flush_fn()
     barrier()
@@ -299,8 +300,8 @@ def main():
             torch.cuda.synchronize()
         probes.append(time.perf_counter() - t1)
     p50_ms = statistics.median(probes) * 1e3 if probes else None
-    if world > 1:
-        e = torch.tensor([p50_ms or 0.0])
+    if world > 1 and p50_ms is not None:
+        e = torch.tensor([p50_ms])
         dist.all_reduce(e, op=dist.ReduceOp.MAX)
         p50_ms = float(e)
 

@@ -1150,7 +1150,7 @@ __device__ __forceinline__ bool mesh_exchange(
     const unsigned long long t0c = __builtin_amdgcn_s_memrealtime();
     while (__hip_atomic_load(&my_mb[lane].seq, __ATOMIC_ACQUIRE,
                              __HIP_MEMORY_SCOPE_SYSTEM) < sq) {
-      if (__builtin_amdgcn_s_memrealtime() - t0c > 25000000ull) {
+      if (__builtin_amdgcn_s_memrealtime() - t0c > 500000000ull) {
         timed_out = true;
         break;
       }

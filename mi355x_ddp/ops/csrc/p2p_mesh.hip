@@ -19,9 +19,11 @@ static hipStream_t cur_stream() {
   return c10::hip::getCurrentHIPStream().stream();
 }
 
-// ~0.25 s at the 100 MHz constant s_memrealtime clock: far beyond any sane
-// xGMI latency, far below anything that looks like a hang to the driver.
-static constexpr unsigned long long kTimeout = 25ull * 1000 * 1000;
+// ~5 s at the 100 MHz constant s_memrealtime clock: generous enough that
+// rank skew (first-launch overheads, allocator warmup) can never trip a
+// false abort, still far below anything that looks like a hang to a
+// watchdog. Steady-state exchanges complete in microseconds.
+static constexpr unsigned long long kTimeout = 500ull * 1000 * 1000;
 
 template <typename T>
 __device__ __forceinline__ float mesh_ldf(const T* p);
