@@ -1208,6 +1208,14 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
   if (lane < K_) ws[lane] = ldf(&param[w_off + lane]);
   if (lane == K_) ws[32] = ldf(&param[b_off]);
   const float inv2B = 2.f / (float)B_;
+  // write-only LDS update: updater lanes keep w in registers (see the
+  // f32 spec kernel)
+  float wreg[KT];
+#pragma unroll
+  for (int tk = 0; tk < KT; ++tk) {
+    const int k = tk * 16 + r;
+    wreg[tk] = (k < K_) ? ldf(&param[w_off + ((k < K_) ? k : 0)]) : 0.f;
+  }
 
   bf16x8 cfa[MT];      // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
   bf16x8 cbb[KT];      // bwd B: X[q*8+j][tk*16+r] (raw)
@@ -1326,10 +1334,12 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
-      if (q == 0 && k < K_)
-        ws[k] = round_store<__hip_bfloat16>(ws[k] - lr * gacc[tk][0]);
-      else if (q == 0 && k == K_)
+      if (q == 0 && k < K_) {
+        wreg[tk] = round_store<__hip_bfloat16>(wreg[tk] - lr * gacc[tk][0]);
+        ws[k] = wreg[tk];
+      } else if (q == 0 && k == K_) {
         ws[32] = round_store<__hip_bfloat16>(bterm - lr * gacc[tk][0]);
+      }
     }
     __syncthreads();  // ws update visible before next iteration's forward
   }
@@ -1361,6 +1371,15 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
   if (lane < K_) ws[lane] = ldf(&param[w_off + lane]);
   if (lane == K_) ws[32] = ldf(&param[b_off]);
   const float inv2B = 2.f / (float)B_;
+  // updater lanes (q==0) also keep their w values in registers so the
+  // end-of-iteration update is WRITE-only to LDS (no dependent LDS read
+  // on the critical path); readers still take w from LDS
+  float wreg[KT];
+#pragma unroll
+  for (int tk = 0; tk < KT; ++tk) {
+    const int k = tk * 16 + r;
+    wreg[tk] = (k < K_) ? ldf(&param[w_off + ((k < K_) ? k : 0)]) : 0.f;
+  }
 
   // Per-lane operand registers, software-pipelined: `c*` hold the step
   // being computed, `n*` receive the next step's loads (issued at the top
@@ -1505,9 +1524,12 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
       const int k = tk * 16 + r;
-      if (q == 0 && k < K_) ws[k] = round_store<T>(ws[k] - lr * gacc[tk][0]);
-      else if (q == 0 && k == K_)
+      if (q == 0 && k < K_) {
+        wreg[tk] = round_store<T>(wreg[tk] - lr * gacc[tk][0]);
+        ws[k] = wreg[tk];
+      } else if (q == 0 && k == K_) {
         ws[32] = round_store<T>(bterm - lr * gacc[tk][0]);
+      }
     }
     __syncthreads();  // ws update visible before next iteration's forward
   }
