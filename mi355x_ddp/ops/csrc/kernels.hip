@@ -127,51 +127,35 @@ __global__ void k_linear_fwd(const T* __restrict__ X, const T* __restrict__ W,
   }
 }
 
-// Split-K + LDS-tiled variant for small-M / large-K shapes (e.g. the
-// ResNet-50 FC, 32x2048 @ 2048x1000): the 4 waves of a workgroup contract
-// DISJOINT K-slices of the SAME 16x16 output tile. Both MFMA operands are
-// strided by K in memory (16 lanes read 16 different rows), so each wave
-// first stages its X/W chunk into LDS with fully COALESCED row loads and
-// feeds the MFMAs from LDS; partials reduce through LDS at the end.
+// Split-K variant for small-M / large-K shapes (e.g. the ResNet-50 FC,
+// 32x2048 @ 2048x1000): the 4 waves of a workgroup contract DISJOINT
+// K-slices of the SAME 16x16 output tile and reduce partials through LDS —
+// 4x the memory-level parallelism of the serial-K kernel on shapes where
+// M-tiling alone cannot fill the chip.
 template <typename T>
 __global__ void k_linear_fwd_splitk(const T* __restrict__ X,
                                     const T* __restrict__ W,
                                     const T* __restrict__ bias,
                                     T* __restrict__ Y, int B, int K, int N) {
-  constexpr int CHUNK = 64;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int tile_m = blockIdx.x;
   const int tile_n = blockIdx.y;
   const int r = lane & 15;
   const int q = lane >> 4;
-  __shared__ float xa[4][16][CHUNK];  // [wave][row][k] — wave-private
-  __shared__ float wb[4][16][CHUNK];
-  __shared__ float red[4][64][4];
-  const int kq = (((K + 3) / 4 + 3) / 4) * 4;  // K-slice per wave
+  const int m = tile_m * 16 + r;
+  const int n = tile_n * 16 + r;
+  const int kq = (((K + 3) / 4 + 3) / 4) * 4;  // K-slice per wave (mult of 4)
   const int k_lo = wave * kq, k_hi = min(K, (wave + 1) * kq);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int kc = k_lo; kc < k_hi; kc += CHUNK) {
-    const int kn = min(CHUNK, k_hi - kc);
-    // coalesced staging: lane l loads column kc+l of each of the 16 rows
 #pragma unroll 4
-    for (int i = 0; i < 16; ++i) {
-      const int m = tile_m * 16 + i;
-      const int n = tile_n * 16 + i;
-      // zero-fill past kn so a ragged tail chunk never feeds stale LDS
-      xa[wave][i][lane] =
-          (lane < kn && m < B) ? ldf(&X[(size_t)m * K + kc + lane]) : 0.f;
-      wb[wave][i][lane] =
-          (lane < kn && n < N) ? ldf(&W[(size_t)n * K + kc + lane]) : 0.f;
-    }
-    // wave-private LDS region: same-wave write->read is ordered by the
-    // compiler's lgkm waits; no block barrier needed
-    for (int k0 = 0; k0 < kn; k0 += 4) {
-      const float a = xa[wave][r][k0 + q];
-      const float b = wb[wave][r][k0 + q];
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
-    }
+  for (int k0 = k_lo; k0 < k_hi; k0 += 4) {
+    const int k = k0 + q;
+    const float a = (m < B && k < K) ? ldf(&X[(size_t)m * K + k]) : 0.f;
+    const float b = (n < N && k < K) ? ldf(&W[(size_t)n * K + k]) : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
+  __shared__ float red[4][64][4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) red[wave][lane][i] = acc[i];
   __syncthreads();
@@ -328,42 +312,31 @@ __global__ void k_linear_bwd_x(const T* __restrict__ dY,
   }
 }
 
-// Split-N + LDS-tiled variant of dX (dX[B,K] = dY[B,N] @ W[N,K]): the
-// dY A-operand is strided by N per lane, so it is staged through LDS
-// with coalesced row loads; the W B-operand is naturally coalesced.
+// Split-N variant of dX (same rationale as k_linear_fwd_splitk: the
+// contraction over N=1000+ is split across the block's 4 waves).
 template <typename T>
 __global__ void k_linear_bwd_x_splitk(const T* __restrict__ dY,
                                       const T* __restrict__ W,
                                       T* __restrict__ dX, int B, int K, int N) {
-  constexpr int CHUNK = 64;
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int tile_m = blockIdx.x;
   const int tile_k = blockIdx.y;
   const int r = lane & 15;
   const int q = lane >> 4;
-  __shared__ float da[4][16][CHUNK];  // [wave][row][j] — wave-private
-  __shared__ float red[4][64][4];
-  const int kc_col = tile_k * 16 + r;
+  const int m = tile_m * 16 + r;
+  const int kc = tile_k * 16 + r;
   const int nq = (((N + 3) / 4 + 3) / 4) * 4;
   const int j_lo = wave * nq, j_hi = min(N, (wave + 1) * nq);
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
-  for (int jc = j_lo; jc < j_hi; jc += CHUNK) {
-    const int jn = min(CHUNK, j_hi - jc);
 #pragma unroll 4
-    for (int i = 0; i < 16; ++i) {
-      const int m = tile_m * 16 + i;
-      da[wave][i][lane] =
-          (lane < jn && m < B) ? ldf(&dY[(size_t)m * N + jc + lane]) : 0.f;
-    }
-    for (int j0 = 0; j0 < jn; j0 += 4) {
-      const int j = jc + j0 + q;
-      const float a = da[wave][r][j0 + q];
-      const float b = (j < N && kc_col < K)
-                          ? ldf(&W[(size_t)j * K + kc_col]) : 0.f;
-      acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
-    }
+  for (int j0 = j_lo; j0 < j_hi; j0 += 4) {
+    const int j = j0 + q;
+    const float a = (m < B && j < N) ? ldf(&dY[(size_t)m * N + j]) : 0.f;
+    const float b = (j < N && kc < K) ? ldf(&W[(size_t)j * K + kc]) : 0.f;
+    acc = __builtin_amdgcn_mfma_f32_16x16x4f32(a, b, acc, 0, 0, 0);
   }
+  __shared__ float red[4][64][4];
 #pragma unroll
   for (int i = 0; i < 4; ++i) red[wave][lane][i] = acc[i];
   __syncthreads();
