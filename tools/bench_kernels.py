@@ -49,10 +49,14 @@ def main():
     xf = torch.rand(32, 2048, device=DEV)
     wf = torch.rand(1000, 2048, device=DEV)
     bf = torch.rand(1000, device=DEV)
-    row("linear fwd 32x2048->1000 (split-K)",
+    row("linear fwd 32x2048->1000 (raw split-K kernel)",
         timeit(lambda: ops.ext().linear_fwd(xf, wf, bf)),
         timeit(lambda: torch.nn.functional.linear(xf, wf, bf)),
         "ResNet FC, rocBLAS ref")
+    row("linear fwd 32x2048->1000 (ops.linear, routed)",
+        timeit(lambda: ops.linear(xf, wf, bf)),
+        timeit(lambda: torch.nn.functional.linear(xf, wf, bf)),
+        "library-shape -> rocBLAS")
 
     # FC dX: 32x1000 @ 1000x2048
     dy = torch.rand(32, 1000, device=DEV)
