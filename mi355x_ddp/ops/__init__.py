@@ -73,10 +73,9 @@ class _HipLinearFn(torch.autograd.Function):
         ctx.has_bias = b is not None
         xc = x.contiguous()
         if _library_gemm_shape(w.shape[1], w.shape[0]):
-            y = torch.mm(xc, w.t())
             if b is not None:
-                y += b
-            return y
+                return torch.addmm(b, xc, w.t())
+            return torch.mm(xc, w.t())
         return ext().linear_fwd(xc, w, b)
 
     @staticmethod
