@@ -2,7 +2,8 @@
 optimizer (§2.3 — DP is the reference's one
 parallelism strategy, re-implemented here from scratch for MI355X)."""
 
-from .comm import ddp_setup, create_comm, GlooComm, RcclCommAdapter  # noqa: F401
+from .comm import (ddp_setup, create_comm, GlooComm,  # noqa: F401
+                   P2pMeshComm, RcclCommAdapter)
 from .reducer import Reducer, Bucket  # noqa: F401
 from .ddp import DDP  # noqa: F401
 from .sgd import FusedSGD  # noqa: F401
