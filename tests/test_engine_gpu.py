@@ -235,3 +235,36 @@ def test_trainer_fast_engine_matches_hooks_on_gpu(tmp_path):
     w_fast = run("persistent")
     assert torch.allclose(w_fast, w_hooks, atol=1e-5), \
         (w_fast - w_hooks).abs().max()
+
+
+def test_generic_multistep_odd_shape_bitwise():
+    # shapes off the (32,20) fast path take the generic rolled kernel —
+    # hold it to the same bitwise-vs-single-step contract (B=48, K=12)
+    from mi355x_ddp import ops
+    from mi355x_ddp.models import toy_model
+
+    def params(seed=7):
+        torch.manual_seed(seed)
+        m = toy_model(12, 1).to(DEV)
+        from mi355x_ddp.parallel.reducer import Reducer
+        red = Reducer(list(m.parameters()), comm=None)
+        b = red.buckets[0]
+        _, widx = red._param_index[m.weight]
+        _, bidx = red._param_index[m.bias]
+        return m, b.flat_param, b.flat_grad, b.offsets[widx], b.offsets[bidx]
+
+    g = torch.Generator().manual_seed(4)
+    X = torch.rand(6 * 48, 12, generator=g).to(DEV)
+    T = torch.rand(6 * 48, 1, generator=g).to(DEV)
+
+    m1, p1, g1, w1, b1 = params()
+    for s in range(6):
+        ops.ext().toy_fused_fwd_bwd(X[s * 48:(s + 1) * 48].contiguous(),
+                                    T[s * 48:(s + 1) * 48].contiguous(),
+                                    p1, g1, torch.Tensor(), True, w1, b1, 0.05)
+    torch.cuda.synchronize()
+
+    m2, p2, g2, w2, b2 = params()
+    ops.ext().toy_multistep(X, T, p2, torch.Tensor(), True, w2, b2, 0.05, 48)
+    torch.cuda.synchronize()
+    assert torch.equal(p1, p2), (p1 - p2).abs().max()
