@@ -113,3 +113,16 @@ def test_sharded_sampler_partition_properties():
         assert list(ours) == list(ref)
 
     check()
+
+
+def test_perm_index_is_bijection_fuzz():
+    from mi355x_ddp.ops import perm_index
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(n=st.integers(2, 5000), seed=st.integers(0, 2**31 - 1))
+    def check(n, seed):
+        seen = {perm_index(seed, p, n) for p in range(n)}
+        assert len(seen) == n and min(seen) == 0 and max(seen) == n - 1
+
+    check()

@@ -187,3 +187,36 @@ def test_persistent_mesh_bf16_world2(tmp_path):
              nprocs=2, join=True)
     w = torch.load(tmp_path / "bf16.pt", weights_only=True)["w"]
     assert torch.isfinite(w).all() and w.abs().sum() > 0
+
+
+def _timeout_worker(rank, world, port, out):
+    from mi355x_ddp import ops
+    _init(rank, world, port)
+    try:
+        from mi355x_ddp.parallel.comm import GlooComm, P2pMeshComm
+        comm = P2pMeshComm(torch.device("cuda", 0), base=GlooComm())
+        if rank == 0:
+            # rank 1 never issues this exchange: the bounded spin must
+            # expire and check() must raise — never a hang
+            t = torch.ones(8, device="cuda")
+            comm._mesh.all_reduce_avg_inline(t)
+            torch.cuda.synchronize()
+            err = None
+            try:
+                comm.check()
+            except RuntimeError as e:
+                err = str(e)
+            assert err is not None and "timed out" in err
+            torch.save({"raised": True}, out)
+        # rank 1 just waits so rank 0's kernel canobserve its absence, then
+        # both leave together
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_mesh_timeout_raises_not_hangs(tmp_path):
+    out = str(tmp_path / "t.pt")
+    mp.spawn(_timeout_worker, args=(2, _free_port(), out), nprocs=2,
+             join=True)
+    assert torch.load(out, weights_only=True)["raised"]
