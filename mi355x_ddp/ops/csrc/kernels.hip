@@ -13,9 +13,15 @@
 // bf16 is the BASELINE.json config-2 capability). Matmul compute uses
 // the exact-f32 MFMA `v_mfma_f32_16x16x4_f32` for f32 storage (no
 // xf32/TF32 exists on gfx950 — cdna_hip_programming.md §3) and
-// `v_mfma_f32_16x16x32_bf16` for bf16 storage in the standalone GEMM; the
-// toy shapes are single-wave and latency-bound, so the MFMA path buys
-// kernel-count and issue-slot economy, not FLOPs.
+// `v_mfma_f32_16x16x32_bf16` for bf16; the toy shapes are single-wave
+// and latency-bound, so the MFMA path buys kernel-count and issue-slot
+// economy, not FLOPs.
+//
+// Beyond parity (design notes in docs/KERNELS.md): the multi-step
+// trainers (`k_toy_multistep_spec`, `k_toy_multistep_bf16w`) run S
+// sequential SGD steps per launch with LDS-resident weights; their MESH
+// variants embed a per-step device-side xGMI all-reduce (p2p_mesh.h);
+// `k_epoch_shard` fuses the epoch shuffle into one copy pass.
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
