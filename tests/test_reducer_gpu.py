@@ -56,3 +56,67 @@ def test_reducer_mlp_matches_autograd(grad_views, cap_mb):
     for p, pr in zip(model.parameters(), ref.parameters()):
         assert torch.allclose(p, pr, atol=1e-5, rtol=1e-4), \
             (p.shape, (p - pr).abs().max())
+
+
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
+def _ddp_gpu_worker(rank, world, port, grad_views, out):
+    import os
+    import torch.multiprocessing  # noqa: F401
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=world)
+    torch.cuda.set_device(0)
+    try:
+        from mi355x_ddp.parallel import DDP, FusedSGD
+        from mi355x_ddp.parallel.comm import GlooComm
+        model = _mlp(0)
+        eng = DDP(model, comm=GlooComm(), bucket_cap_mb=0.5,
+                  grad_views=grad_views)
+        opt = FusedSGD(model.parameters(), lr=0.01)
+        opt.attach_reducer(eng.reducer)
+        g = torch.Generator().manual_seed(50 + rank)  # different data/rank
+        x = torch.randn(8, 64, generator=g).to(DEV)
+        t = torch.randn(8, 10, generator=g).to(DEV)
+        for _ in range(4):
+            torch.nn.functional.mse_loss(eng(x), t).backward()
+            eng.finalize_backward()
+            opt.step()
+        torch.cuda.synchronize()
+        if rank == 0:
+            torch.save([p.detach().cpu() for p in model.parameters()], out)
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.parametrize("grad_views", [True, False])
+def test_ddp_multibucket_world2_on_device(grad_views, tmp_path):
+    # distributed multi-bucket path with DEVICE tensors: 2 processes share
+    # one GPU, gradient buckets all-reduced over the comm layer == a
+    # single process training on the concatenated batch
+    import torch.multiprocessing as mp
+    out = str(tmp_path / "ddp.pt")
+    mp.spawn(_ddp_gpu_worker, args=(2, _free_port(), grad_views, out),
+             nprocs=2, join=True)
+    got = torch.load(out, weights_only=True)
+
+    # reference: single process, full batch (grad averaging == mean loss
+    # over the union for equal-sized shards)
+    ref = _mlp(0)
+    opt = torch.optim.SGD(ref.parameters(), lr=0.01)
+    gs = [torch.Generator().manual_seed(50 + r) for r in range(2)]
+    x = torch.cat([torch.randn(8, 64, generator=g) for g in gs]).to(DEV)
+    t = torch.cat([torch.randn(8, 10, generator=g) for g in gs]).to(DEV)
+    for _ in range(4):
+        opt.zero_grad()
+        torch.nn.functional.mse_loss(ref(x), t).backward()
+        opt.step()
+    for p, pr in zip(got, ref.parameters()):
+        assert torch.allclose(p, pr.detach().cpu(), atol=1e-5, rtol=1e-4), \
+            (p - pr.detach().cpu()).abs().max()
