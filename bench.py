@@ -47,6 +47,11 @@ def parse_args():
     p.add_argument("--dtype", choices=["fp32", "bf16"], default="fp32",
                    help="compute dtype (fp32 = the reference's; bf16 = "
                         "BASELINE.json config 2)")
+    p.add_argument("--loss", choices=["mse", "ce"], default="mse",
+                   help="mse (default: the reference's multinode loss; "
+                        "actually trains) or ce (the reference's stage-1 "
+                        "loss — degenerate over one logit: zero gradients, "
+                        "SURVEY §2.1)")
     return p.parse_args()
 
 
@@ -141,7 +146,7 @@ class DeviceData:
         return self._views[s]
 
 
-def build_engine(kind, comm, lr, device, dtype=torch.float32):
+def build_engine(kind, comm, lr, device, dtype=torch.float32, use_mse=True):
     from mi355x_ddp.engine import (GraphedToyStep, PersistentToyStep,
                                    ToyFusedStep)
     from mi355x_ddp.models import toy_model
@@ -157,8 +162,10 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
         opt = FusedSGD(model.parameters(), lr=lr)
         opt.attach_reducer(engine.reducer)
 
+        loss_fn = ops.mse_loss if use_mse else ops.cross_entropy
+
         def step(x, t):
-            loss = ops.mse_loss(engine(x), t)
+            loss = loss_fn(engine(x), t)
             loss.backward()
             engine.finalize_backward()
             opt.step()
@@ -169,7 +176,7 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32):
         kind = "fused"  # without a mesh comm, multi-step is world-1 only
     cls = {"persistent": PersistentToyStep, "graph": GraphedToyStep,
            "fused": ToyFusedStep}[kind]
-    eng = cls(model, comm=comm, lr=lr, use_mse=True)
+    eng = cls(model, comm=comm, lr=lr, use_mse=use_mse)
     if comm is not None:
         eng.reducer.broadcast_params(root=0)
     return eng.step, getattr(eng, "flush", noflush), eng
@@ -229,7 +236,7 @@ def main():
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
     step_fn, flush_fn, engine_obj = build_engine(
         args.engine if use_cuda else "autograd", comm, args.lr, device,
-        dtype if use_cuda else torch.float32)
+        dtype if use_cuda else torch.float32, use_mse=args.loss == "mse")
     data = DeviceData(args.dataset, rank, world, args.batch, device,
                       dtype=dtype if use_cuda else torch.float32)
 
@@ -333,7 +340,7 @@ def main():
                             "GraphedToyStep": "graph",
                             "ToyFusedStep": "fused"}[
                                type(engine_obj).__name__]),
-                "loss": "mse",
+                "loss": args.loss,
                 "p50_step_ms": p50_ms,
             },
         }
