@@ -149,8 +149,17 @@ class Trainer:
                 and len(self.optimizer.param_groups) == 1):
             return False
         comm = create_comm(self.device)
+        if comm is not None:
+            # try the device-side mesh so the multi-step engine also works
+            # at world > 1 (validated vs gloo; falls back on any failure)
+            try:
+                from .parallel.comm import P2pMeshComm
+                comm = P2pMeshComm(self.device, base=comm)
+            except Exception as e:
+                print(f"[mi355x_ddp] mesh unavailable ({e!r}); "
+                      "fast engine uses the base transport", flush=True)
         if kind == "auto":
-            kind = "persistent" if comm is None else "fused"
+            kind = "persistent"
         from .engine import PersistentToyStep, ToyFusedStep
         lr = self.optimizer.param_groups[0]["lr"]
         if kind == "persistent":

@@ -220,3 +220,40 @@ def test_mesh_timeout_raises_not_hangs(tmp_path):
     mp.spawn(_timeout_worker, args=(2, _free_port(), out), nprocs=2,
              join=True)
     assert torch.load(out, weights_only=True)["raised"]
+
+
+def _trainer_mesh_worker(rank, world, port, out_dir):
+    from mi355x_ddp.data import ToyDataset, prepare_dataloader
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.trainer import Trainer
+    _init(rank, world, port)
+    try:
+        torch.manual_seed(2)
+        model = toy_model(20, 1)
+        loader = prepare_dataloader(ToyDataset(256, seed=9), 32,
+                                    distributed=True, shuffle=False,
+                                    num_replicas=world, rank=rank)
+        opt = FusedSGD(model.parameters(), lr=0.05)
+        tr = Trainer(model, loader, opt, 0, save_every=10**9, loss_fn="mse",
+                     engine="persistent",
+                     checkpoint_path=os.path.join(out_dir, "c.pt"))
+        assert tr._engine is not None
+        assert getattr(tr._engine, "_mesh", None) is not None
+        tr.train(2)
+        torch.cuda.synchronize()
+        if rank == 0:
+            torch.save({"w": model.weight.detach().cpu()},
+                       os.path.join(out_dir, "trainer_mesh.pt"))
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_trainer_persistent_mesh_world2(tmp_path):
+    # the public Trainer API drives the multi-step + in-kernel-mesh engine
+    # at world 2 (one device)
+    mp.spawn(_trainer_mesh_worker, args=(2, _free_port(), str(tmp_path)),
+             nprocs=2, join=True)
+    w = torch.load(tmp_path / "trainer_mesh.pt", weights_only=True)["w"]
+    assert torch.isfinite(w).all() and w.abs().sum() > 0
