@@ -148,7 +148,13 @@ class Trainer:
                 and isinstance(self.optimizer, FusedSGD)
                 and len(self.optimizer.param_groups) == 1):
             return False
-        comm = create_comm(self.device)
+        from .parallel.comm import GlooComm
+        try:
+            comm = create_comm(self.device)
+        except Exception as e:  # e.g. RCCL refusing a shared test device
+            print(f"[mi355x_ddp] native comm init failed ({e!r}); "
+                  "fast engine falls back to gloo transport", flush=True)
+            comm = GlooComm()
         if comm is not None:
             # try the device-side mesh so the multi-step engine also works
             # at world > 1 (validated vs gloo; falls back on any failure)
