@@ -201,28 +201,15 @@ def main():
     if world > 1:
         # gloo for rendezvous/barriers only; gradient bytes ride RcclComm
         dist.init_process_group("gloo", rank=rank, world_size=world)
-        from mi355x_ddp.parallel.comm import (GlooComm, P2pMeshComm,
-                                              RcclCommAdapter)
+        from mi355x_ddp.parallel.comm import GlooComm, build_gpu_comm
         if use_cuda:
-            try:
-                comm = RcclCommAdapter(device)
-                comm_kind = "rccl"
-            except Exception as e:  # keep the scale run alive, but say so
-                print(f"[bench] RcclComm init FAILED ({e!r}); "
-                      "falling back to gloo transport — numbers are NOT "
-                      "the native RCCL path", file=sys.stderr, flush=True)
-                comm = GlooComm()
-                comm_kind = "gloo-fallback"
-            if os.environ.get("MI355X_P2P", "1") != "0":
-                # device-side xGMI mesh for the 84 B all-reduce; validated
-                # against gloo at setup, raises (never hangs) on timeout
-                try:
-                    comm = P2pMeshComm(device, base=comm)
-                    comm_kind = "p2p-mesh+" + comm_kind
-                except Exception as e:
-                    print(f"[bench] P2pMesh unavailable ({e!r}); "
-                          f"staying on {comm_kind}", file=sys.stderr,
-                          flush=True)
+            # hang-safe transport ladder: RCCL (-> gloo), + the xGMI mesh
+            # when every rank agrees it set up AND cross-validated; any
+            # downgrade happens on ALL ranks together so collectives never
+            # mismatch (mi355x_ddp.parallel.comm.build_gpu_comm)
+            comm, comm_kind = build_gpu_comm(
+                device, want_mesh=os.environ.get("MI355X_P2P", "1") != "0",
+                log=lambda m: print(m, file=sys.stderr, flush=True))
         else:
             comm = GlooComm()
             comm_kind = "gloo-cpu"

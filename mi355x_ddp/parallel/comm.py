@@ -128,7 +128,7 @@ class P2pMeshComm:
 
     _KEY = "mi355x_ddp/p2p_mesh"
 
-    def __init__(self, device: torch.device, base):
+    def __init__(self, device: torch.device, base, validate: bool = True):
         from .. import ops
         assert dist.is_initialized()
         self.base = base
@@ -144,7 +144,8 @@ class P2pMeshComm:
             handles.append(base64.b64decode(store.get(f"{self._KEY}/{r}")))
         self._mesh.connect(handles)
         self._device = device
-        self.validate()
+        if validate:
+            self.validate()
 
     def validate(self) -> None:
         """One mesh all-reduce cross-checked against the gloo group."""
@@ -182,6 +183,61 @@ class P2pMeshComm:
 
     def barrier(self) -> None:
         self.base.barrier()
+
+
+def all_ranks_agree(ok: bool) -> bool:
+    """World-wide MIN over a per-rank success flag (gloo/default group).
+
+    Per-rank try/except around transport setup is NOT enough: if only
+    SOME ranks fall back, the world ends up issuing mismatched collectives
+    on different transports, which HANGS instead of failing. Every
+    transport decision must therefore be agreed by all ranks — call this
+    at the same program point on every rank."""
+    t = torch.tensor([1 if ok else 0])
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    return bool(t.item())
+
+
+def build_gpu_comm(device: torch.device, want_mesh: bool = True,
+                   log=print):
+    """The full, hang-safe transport ladder for a GPU world:
+    RCCL -> (gloo fallback), then optionally the xGMI mesh layered on top
+    — with EVERY step agreed across ranks (see all_ranks_agree) and the
+    mesh cross-validated against gloo before adoption.
+    Returns (comm, kind_string)."""
+    assert dist.is_initialized()
+    comm = None
+    kind = "rccl"
+    try:
+        comm = RcclCommAdapter(device)
+    except Exception as e:
+        log(f"[mi355x_ddp] RcclComm init failed on rank "
+            f"{dist.get_rank()} ({e!r})")
+        kind = "gloo-fallback"
+    if not all_ranks_agree(comm is not None):
+        comm = GlooComm()
+        kind = "gloo-fallback"
+    if comm is None:  # this rank failed but (impossible here) others agreed
+        comm = GlooComm()
+    if want_mesh:
+        mesh = None
+        try:
+            mesh = P2pMeshComm(device, base=comm, validate=False)
+        except Exception as e:
+            log(f"[mi355x_ddp] P2pMesh setup failed on rank "
+                f"{dist.get_rank()} ({e!r})")
+        if all_ranks_agree(mesh is not None):
+            ok = True
+            try:
+                mesh.validate()
+            except Exception as e:
+                log(f"[mi355x_ddp] P2pMesh validation failed on rank "
+                    f"{dist.get_rank()} ({e!r})")
+                ok = False
+            if all_ranks_agree(ok):
+                return mesh, "p2p-mesh+" + kind
+        log(f"[mi355x_ddp] mesh not adopted; staying on {kind}")
+    return comm, kind
 
 
 def create_comm(device: torch.device):

@@ -140,7 +140,6 @@ class Trainer:
         import torch.distributed as dist
         from .models.toy import HipLinear
         from .parallel import FusedSGD
-        from .parallel.comm import create_comm
         m = self.model
         if not (isinstance(m, HipLinear) and m.out_features == 1
                 and self.device.type == "cuda"
@@ -148,22 +147,14 @@ class Trainer:
                 and isinstance(self.optimizer, FusedSGD)
                 and len(self.optimizer.param_groups) == 1):
             return False
-        from .parallel.comm import GlooComm
-        try:
-            comm = create_comm(self.device)
-        except Exception as e:  # e.g. RCCL refusing a shared test device
-            print(f"[mi355x_ddp] native comm init failed ({e!r}); "
-                  "fast engine falls back to gloo transport", flush=True)
-            comm = GlooComm()
-        if comm is not None:
-            # try the device-side mesh so the multi-step engine also works
-            # at world > 1 (validated vs gloo; falls back on any failure)
-            try:
-                from .parallel.comm import P2pMeshComm
-                comm = P2pMeshComm(self.device, base=comm)
-            except Exception as e:
-                print(f"[mi355x_ddp] mesh unavailable ({e!r}); "
-                      "fast engine uses the base transport", flush=True)
+        if dist.is_available() and dist.is_initialized() \
+                and dist.get_world_size() > 1:
+            # hang-safe, world-agreed transport ladder (RCCL -> gloo, +
+            # mesh only when every rank set it up AND it cross-validated)
+            from .parallel.comm import build_gpu_comm
+            comm, _kind = build_gpu_comm(self.device)
+        else:
+            comm = None
         if kind == "auto":
             kind = "persistent"
         from .engine import PersistentToyStep, ToyFusedStep
