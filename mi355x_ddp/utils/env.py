@@ -13,6 +13,13 @@ MI355X_EPOCH_SIDE_STREAM  1 = gather epoch shards on a side stream
                         (measured slower; default 0)
 MI355X_PREFETCH         0 = disable the Trainer's copy-stream H2D
                         prefetcher (default on for GPU)
+MI355X_P2P              0 = disable the xGMI mesh all-reduce in bench.py
+                        (default on; mesh is only adopted after all ranks
+                        agree it set up and cross-validated)
+
+RCCL's own tuning envs (NCCL_ALGO, NCCL_PROTO, NCCL_MIN/MAX_NCHANNELS)
+pass straight through to the large-bucket collective path — the knobs
+SURVEY §5.8 names for per-size algorithm selection over xGMI.
 """
 
 import os
