@@ -268,3 +268,58 @@ def test_generic_multistep_odd_shape_bitwise():
     ops.ext().toy_multistep(X, T, p2, torch.Tensor(), True, w2, b2, 0.05, 48)
     torch.cuda.synchronize()
     assert torch.equal(p1, p2), (p1 - p2).abs().max()
+
+
+def test_persistent_deferral_state_machine_fuzz():
+    # random interleavings of step()/step_shard()/flush() must execute the
+    # submitted batches in submission order — bitwise vs the eager engine
+    import random
+    from mi355x_ddp.engine import PersistentToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+
+    rng = random.Random(1234)
+    g = torch.Generator().manual_seed(77)
+    Xs = torch.rand(64 * 32, 20, generator=g).to(DEV)
+    Ts = torch.rand(64 * 32, 1, generator=g).to(DEV)
+    pool = torch.rand(32 * 32, 20, generator=g).to(DEV)
+    poolT = torch.rand(32 * 32, 1, generator=g).to(DEV)
+
+    for trial in range(4):
+        torch.manual_seed(9 + trial)
+        m_ref = toy_model(20, 1).to(DEV)
+        eager = ToyFusedStep(m_ref, comm=None, lr=0.03, use_mse=True)
+        torch.manual_seed(9 + trial)
+        m = toy_model(20, 1).to(DEV)
+        eng = PersistentToyStep(m, comm=None, lr=0.03, use_mse=True)
+        eng.bind_shard(Xs, Ts, 32)
+        submitted = []
+        next_seq = 0
+        for _ in range(60):
+            r = rng.random()
+            if r < 0.45:  # sequential shard step (the common fast path)
+                i = next_seq % 64
+                next_seq += 1
+                eng.step_shard(i)
+                submitted.append((Xs[i * 32:(i + 1) * 32],
+                                  Ts[i * 32:(i + 1) * 32]))
+            elif r < 0.65:  # random shard index (breaks the run)
+                i = rng.randrange(64)
+                next_seq = i + 1
+                eng.step_shard(i)
+                submitted.append((Xs[i * 32:(i + 1) * 32],
+                                  Ts[i * 32:(i + 1) * 32]))
+            elif r < 0.9:  # tensor-API step from a different buffer
+                j = rng.randrange(32)
+                x, t = pool[j * 32:(j + 1) * 32], poolT[j * 32:(j + 1) * 32]
+                eng.step(x, t)
+                submitted.append((x, t))
+            else:
+                eng.flush()
+        eng.flush()
+        torch.cuda.synchronize()
+        for x, t in submitted:
+            eager.step(x.contiguous(), t.contiguous())
+        torch.cuda.synchronize()
+        assert torch.equal(m.weight.detach(), m_ref.weight.detach()), \
+            (trial, (m.weight - m_ref.weight).abs().max())
+        assert torch.equal(m.bias.detach(), m_ref.bias.detach())
