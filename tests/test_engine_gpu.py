@@ -323,3 +323,32 @@ def test_persistent_deferral_state_machine_fuzz():
         assert torch.equal(m.weight.detach(), m_ref.weight.detach()), \
             (trial, (m.weight - m_ref.weight).abs().max())
         assert torch.equal(m.bias.detach(), m_ref.bias.detach())
+
+
+def test_spec_multistep_batch64_bitwise():
+    # the batch-64 instantiation of the fast-path kernel (246 VGPRs, no
+    # spills) == repeated single-step launches, bitwise
+    from mi355x_ddp.engine import PersistentToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    g = torch.Generator().manual_seed(21)
+    Xf = torch.rand(10 * 64, 20, generator=g).to(DEV)
+    Tf = torch.rand(10 * 64, 1, generator=g).to(DEV)
+
+    torch.manual_seed(5)
+    m_ref = toy_model(20, 1).to(DEV)
+    eager = ToyFusedStep(m_ref, comm=None, lr=0.04, use_mse=True)
+    for s in range(10):
+        eager.step(Xf[s * 64:(s + 1) * 64].contiguous(),
+                   Tf[s * 64:(s + 1) * 64].contiguous())
+    torch.cuda.synchronize()
+
+    torch.manual_seed(5)
+    m = toy_model(20, 1).to(DEV)
+    eng = PersistentToyStep(m, comm=None, lr=0.04, use_mse=True)
+    eng.bind_shard(Xf, Tf, 64)
+    for s in range(10):
+        eng.step_shard(s)
+    eng.flush()
+    torch.cuda.synchronize()
+    assert torch.equal(m.weight.detach(), m_ref.weight.detach())
+    assert torch.equal(m.bias.detach(), m_ref.bias.detach())
