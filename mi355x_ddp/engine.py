@@ -188,7 +188,7 @@ class PersistentToyStep(ToyFusedStep):
                 x0 = t0 = None
                 count = 0
                 if n == 1:
-                    eager(xx, tt)
+                    launch(xx, tt, xx.shape[0])  # S=1: bitwise == fused
                 else:
                     B, K = xx.shape
                     xall = xx.as_strided((n * B, K), (K, 1))
