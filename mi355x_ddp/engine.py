@@ -3,20 +3,20 @@
 The reference hot loop (single_gpu.py:21-26 / SURVEY §3.5) costs, per step:
 zero_grad + fwd GEMM + loss fwd + loss bwd + two bwd GEMMs + bucket
 all-reduce + SGD — with 84 B of gradients the whole thing is launch-latency
-bound (SURVEY §7 hard-part 2). `ToyFusedStep` collapses it to:
+bound (SURVEY §7 hard-part 2). Three engines, in increasing aggression:
 
-    1 kernel  (fused fwd + loss grad + bwd, writes grads into the bucket)
-  [ + 1 RCCL all-reduce when world > 1 ]
-    1 kernel  (fused SGD over the flat bucket, zeroes grads)
+- `ToyFusedStep`: ONE kernel (fused fwd + loss grad + bwd [+ in-kernel SGD
+  at world 1]); at world > 1, + one tiny all-reduce + one fused-SGD kernel.
+- `GraphedToyStep`: the fused step captured in a hipGraph and replayed.
+- `PersistentToyStep` (the bench default): DEFERRED runs of consecutive
+  steps execute as one multi-step kernel with the weights LDS-resident;
+  at world > 1 with a `P2pMeshComm` the kernel performs an in-kernel xGMI
+  mesh all-reduce per step. See docs/KERNELS.md.
 
-`GraphedToyStep` additionally captures the whole step (including the RCCL
-collective) into a hipGraph once and replays it per step — HIP streams and
-graphs instead of per-launch host dispatch.
-
-Both engines run the SAME model parameters the generic autograd path
+All engines train the SAME model parameters the generic autograd path
 trains: the Reducer's flat bucket is shared state, so checkpoints and the
-Trainer API are unaffected. Numerics equivalence with the autograd path is
-covered by tests/test_engine_gpu.py.
+Trainer API are unaffected. Equivalence with the autograd path (and the
+bitwise multi==single contract) is covered by tests/test_engine_gpu.py.
 """
 
 from __future__ import annotations
