@@ -111,3 +111,26 @@ def test_trainer_fast_engine_falls_back_on_cpu(tmp_path):
     w0 = model.weight.detach().clone()
     tr.train(1)
     assert not torch.equal(model.weight.detach(), w0)
+
+
+def test_reference_written_snapshot_loads(tmp_chdir):
+    # a snapshot produced by the REFERENCE's literal code shape
+    # (torch.nn.Linear state_dict under MODEL_STATE, multigpu_torchrun.py:57-62)
+    # restores into this framework's Trainer/HipLinear unchanged
+    import torch
+    from mi355x_ddp.data import ToyDataset, prepare_dataloader
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.trainer import Trainer
+
+    ref_model = torch.nn.Linear(20, 1)
+    torch.save({"MODEL_STATE": ref_model.state_dict(), "EPOCHS_RUN": 5},
+               "snapshot.pt")
+
+    model = toy_model(20, 1)
+    loader = prepare_dataloader(ToyDataset(64, seed=0), 32, shuffle=False)
+    opt = torch.optim.SGD(model.parameters(), lr=1e-3)
+    tr = Trainer(model, loader, opt, "cpu", save_every=10**9, loss_fn="mse",
+                 wrap_ddp=False, snapshot_path="snapshot.pt")
+    assert tr.epochs_run == 5
+    assert torch.equal(model.weight.detach(), ref_model.weight.detach())
+    assert torch.equal(model.bias.detach(), ref_model.bias.detach())
