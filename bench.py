@@ -221,17 +221,25 @@ def main():
             torch.cuda.synchronize()
 
     dtype = torch.bfloat16 if args.dtype == "bf16" else torch.float32
-    step_fn, flush_fn, engine_obj = build_engine(
-        args.engine if use_cuda else "autograd", comm, args.lr, device,
-        dtype if use_cuda else torch.float32, use_mse=args.loss == "mse")
     data = DeviceData(args.dataset, rank, world, args.batch, device,
                       dtype=dtype if use_cuda else torch.float32)
-
-    # shard-bound fast path: bind each epoch's device-resident shard once,
-    # then drive the engine by batch INDEX (no per-step view construction)
-    eng_bind = getattr(engine_obj, "bind_shard", None)
-    eng_step_shard = getattr(engine_obj, "step_shard", None)
     spe = data.steps_per_epoch
+
+    step_fn = flush_fn = engine_obj = None
+    eng_bind = eng_step_shard = None
+
+    def make_engine():
+        nonlocal step_fn, flush_fn, engine_obj, eng_bind, eng_step_shard
+        step_fn, flush_fn, engine_obj = build_engine(
+            args.engine if use_cuda else "autograd", comm, args.lr, device,
+            dtype if use_cuda else torch.float32, use_mse=args.loss == "mse")
+        # shard-bound fast path: bind each epoch's shard once, then drive
+        # the engine by batch INDEX (no per-step view construction)
+        eng_bind = getattr(engine_obj, "bind_shard", None)
+        eng_step_shard = getattr(engine_obj, "step_shard", None)
+        data.bound_epoch = -1
+
+    make_engine()
 
     def run_steps(start, n):
         if eng_bind is None:
@@ -258,10 +266,35 @@ def main():
     run_steps(0, args.warmup)
     flush_fn()
     barrier()
-    comm_check()
-    if comm is not None and hasattr(comm, "validate"):
-        comm.validate()  # mesh transport re-verified against gloo after
-        # warmup — catches coherence problems before anything is timed
+    # Post-warmup transport health check. If the mesh misbehaved during
+    # warmup on ANY rank (timeout / cross-validation mismatch), EVERY rank
+    # downgrades to the base transport together (mismatched transports
+    # would hang) and warmup is redone on the safe path.
+    ok = True
+    try:
+        comm_check()
+        if comm is not None and hasattr(comm, "validate"):
+            comm.validate()
+    except Exception as e:
+        print(f"[bench] transport health check failed: {e!r}",
+              file=sys.stderr, flush=True)
+        ok = False
+    if os.environ.get("MI355X_TEST_FAIL_HEALTH") == "1":
+        ok = False  # test-only: exercise the downgrade + re-warmup path
+    if world > 1:
+        from mi355x_ddp.parallel.comm import all_ranks_agree
+        if not all_ranks_agree(ok):
+            if hasattr(comm, "base"):
+                comm = comm.base
+                comm_kind = comm_kind.split("+", 1)[-1] + "(post-warmup)"
+                print(f"[bench] downgraded to {comm_kind}; redoing warmup",
+                      file=sys.stderr, flush=True)
+            make_engine()
+            run_steps(0, args.warmup)
+            flush_fn()
+            barrier()
+    elif not ok:
+        raise RuntimeError("transport health check failed at world 1")
 
     # -- timed region: exactly K steps (any deferred launches are flushed
     #    INSIDE the bracket — all K steps' work executes before the
