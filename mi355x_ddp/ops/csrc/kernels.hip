@@ -1164,7 +1164,9 @@ __device__ __forceinline__ bool mesh_exchange(
     }
   }
   if (__any(timed_out)) return false;
-  __threadfence();
+  // system-scope fence: subsequent data reads (by OTHER lanes than the
+  // acquirers) must observe the remote ranks' mailbox writes
+  __threadfence_system();
   __syncthreads();
   if (q == 0) {
 #pragma unroll

@@ -81,7 +81,7 @@ k_mesh_allreduce(T* __restrict__ grad, int n, unsigned long long seq,
     if (t == 0) *err_host = 1u;
     return;  // grad left as-is; host check() raises before results are used
   }
-  __threadfence();  // order the acquired slots' data reads below
+  __threadfence_system();  // remote mailbox writes visible to all lanes
 
   if (t < n) {
     float sum = 0.f;
