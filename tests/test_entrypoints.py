@@ -77,7 +77,7 @@ def test_torchrun_elastic_restart_after_crash(tmp_path):
     # so one clean retry with a shorter deadline keeps the test honest
     # without inheriting the agent's flakiness.
     r = None
-    for attempt in range(2):
+    for attempt in range(3):
         for f in (tmp_path / "snapshot.pt", marker):
             if f.exists():
                 f.unlink()
@@ -89,7 +89,7 @@ def test_torchrun_elastic_restart_after_crash(tmp_path):
                      env_extra={"MI355X_FAULT_EPOCH": "1",
                                 "MI355X_FAULT_RANK": "1",
                                 "MI355X_FAULT_ONCE_FILE": str(marker)},
-                     timeout=240)
+                     timeout=200)
         except subprocess.TimeoutExpired:
             continue
         if r.returncode == 0:
