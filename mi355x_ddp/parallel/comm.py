@@ -42,6 +42,22 @@ def ddp_setup(rank: Optional[int] = None, world_size: Optional[int] = None,
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "12355")
         dist.init_process_group(backend, rank=rank, world_size=world_size)
+    elif "TORCHELASTIC_RESTART_COUNT" in os.environ:
+        # Restart-safety: under torchrun the agent-hosted TCPStore SURVIVES
+        # worker restarts, so plain env:// init can read a crashed peer's
+        # STALE transport address from the previous attempt (a race — the
+        # new peer may or may not have overwritten the key yet), failing
+        # with connection-refused or wedging a rank inside native connect.
+        # Keying the process-group store by the elastic attempt number
+        # gives every restart a clean namespace.
+        attempt = os.environ["TORCHELASTIC_RESTART_COUNT"]
+        r = int(os.environ["RANK"])
+        w = int(os.environ["WORLD_SIZE"])
+        store = dist.TCPStore(os.environ["MASTER_ADDR"],
+                              int(os.environ["MASTER_PORT"]),
+                              w, is_master=False)
+        store = dist.PrefixStore(f"mi355x/attempt_{attempt}", store)
+        dist.init_process_group(backend, store=store, rank=r, world_size=w)
     else:
         dist.init_process_group(backend)
     if torch.cuda.is_available():
