@@ -73,6 +73,17 @@ def _agree_worker(rank, world, port, out_dir):
         # EVERY rank must have downgraded together — a split would hang
         assert kind == "gloo-fallback", kind
         assert isinstance(c, comm_mod.GlooComm)
+
+        # want_mesh on a CPU host: P2pMesh setup fails on every rank and
+        # the world agrees to stay on the base transport
+        comm_mod.RcclCommAdapter = FakeRccl
+        try:
+            c2, kind2 = comm_mod.build_gpu_comm(torch.device("cpu"),
+                                                want_mesh=True,
+                                                log=lambda m: None)
+        finally:
+            comm_mod.RcclCommAdapter = orig
+        assert kind2 == "gloo-fallback" and isinstance(c2, comm_mod.GlooComm)
         torch.save({"kind": kind},
                    os.path.join(out_dir, f"agree{rank}.pt"))
         torch.distributed.barrier()
