@@ -1225,11 +1225,20 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
     wreg[tk] = (k < K_) ? ldf(&param[w_off + ((k < K_) ? k : 0)]) : 0.f;
   }
 
-  bf16x8 cfa[MT];      // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
-  bf16x8 cbb[KT];      // bwd B: X[q*8+j][tk*16+r] (raw)
-  float ctR[MT][4];    // targets for rows tm*16+q*4+i
+  // Two operand register sets, DISTANCE-2 prefetch: iteration s consumes
+  // set s%2 and — at the same late, off-critical-path point as before —
+  // issues the loads for step s+2 into that same set. The vmcnt wait for
+  // a set therefore lands a FULL iteration after its loads issued (no
+  // adopt-copies, no issue at the chain head — the two failure modes of
+  // the earlier pipelining attempts, kept in git history).
+  struct BSet {
+    bf16x8 fa[MT];     // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
+    bf16x8 bb[KT];     // bwd B: X[q*8+j][tk*16+r] (raw)
+    float tR[MT][4];   // targets for rows tm*16+q*4+i
+  };
+  BSet setA, setB;
 
-  auto prefetch = [&](int s) {
+  auto prefetch = [&](BSet& R, int s) {
     const __hip_bfloat16* Xs = X + (size_t)s * (B_ * K_);
     const __hip_bfloat16* Ts = Tg + (size_t)s * B_;
 #pragma unroll
@@ -1239,7 +1248,7 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
       for (int j = 0; j < 8; ++j) {
         const int k = q * 8 + j;
         const int kc = (k < K_) ? k : K_ - 1;
-        cfa[tm][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)m * K_ + kc]);
+        R.fa[tm][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)m * K_ + kc]);
       }
     }
 #pragma unroll
@@ -1249,19 +1258,23 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
         const int i = q * 8 + j;  // < 32 == B_
-        cbb[tk][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)i * K_ + kc]);
+        R.bb[tk][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)i * K_ + kc]);
       }
     }
 #pragma unroll
     for (int tm = 0; tm < MT; ++tm)
 #pragma unroll
       for (int i = 0; i < 4; ++i)
-        ctR[tm][i] = ldf(&Ts[tm * 16 + q * 4 + i]);
+        R.tR[tm][i] = ldf(&Ts[tm * 16 + q * 4 + i]);
   };
 
-  prefetch(0);
+  prefetch(setA, 0);
+  if (S > 1) prefetch(setB, 1);
   float loss_last = 0.f;
-  for (int s = 0; s < S; ++s) {
+  auto body = [&](BSet& R, int s) {
+    const bf16x8* cfa = R.fa;
+    const bf16x8* cbb = R.bb;
+    const auto& ctR = R.tR;
     // forward B-operand: w as bf16 (stored values round-trip bf16 exactly),
     // zero-padded for k >= K_ so raw garbage in A contributes nothing
     bf16x8 b8{};
@@ -1321,7 +1334,7 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
                                                          gacc[tk], 0, 0, 0);
     }
 
-    if (s + 1 < S) prefetch(s + 1);  // operand regs dead from here
+    if (s + 2 < S) prefetch(R, s + 2);  // this set's regs dead from here
 
     if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
@@ -1333,7 +1346,7 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
                              mworld, minv_world,
                              seq0 + (unsigned long long)s)) {
         if (lane == 0) *mesh_err = 1u;
-        return;
+        return false;
       }
 #pragma unroll
       for (int tk = 0; tk < KT; ++tk) gacc[tk][0] = gval[tk];
@@ -1350,7 +1363,15 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
       }
     }
     __syncthreads();  // ws update visible before next iteration's forward
+    return true;
+  };
+
+  int s = 0;
+  for (; s + 2 <= S; s += 2) {
+    if (!body(setA, s)) return;
+    if (!body(setB, s + 1)) return;
   }
+  if (s < S && !body(setA, s)) return;
 
   if (lane < K_) stf(&param[w_off + lane], ws[lane]);
   if (lane == K_) stf(&param[b_off], ws[32]);
