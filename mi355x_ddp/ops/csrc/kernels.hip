@@ -1410,22 +1410,17 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
     wreg[tk] = (k < K_) ? ldf(&param[w_off + ((k < K_) ? k : 0)]) : 0.f;
   }
 
-  // Per-lane operand registers. Loads are raw (address-clamped, no value
-  // select — ternary-guarded loads compile to serialized exec-masked
-  // bursts). For B_<=32 there are TWO sets with DISTANCE-2 prefetch:
-  // iteration s consumes set s%2 and, at the late off-critical-path point
-  // after the backward MFMAs, issues step s+2's loads into that same set
-  // — a full iteration of latency shadow with no copies. Larger B_ would
-  // overflow the register file with two sets and keeps distance 1.
-  constexpr int NSETS = (B_ <= 32) ? 2 : 1;
-  struct FSet {
-    float fA[MT][KS];  // fwd A: X[tm*16+r][4*kk+q]
-    float bB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
-    float tR[MT][4];   // targets for rows tm*16+q*4+i
-  };
-  FSet setA, setB;
+  // Per-lane operand registers, software-pipelined: `c*` hold the step
+  // being computed, `n*` receive the next step's loads (issued at the top
+  // of the iteration, consumed — behind one vmcnt wait — at its end).
+  // All loads are UNCONDITIONAL with clamped addresses + value selects:
+  // ternary-guarded loads compile to exec-masked branch-per-element code
+  // that serializes the burst (seen in the r01b ISA), selects do not.
+  float cfA[MT][KS];  // fwd A: X[tm*16+r][4*kk+q]
+  float cbB[KT][BS];  // bwd B: X[4*ii+q][tk*16+r]
+  float ctR[MT][4];   // targets for rows tm*16+q*4+i
 
-  auto prefetch = [&](FSet& R, int s) {
+  auto prefetch = [&](int s) {
     const T* Xs = X + (size_t)s * (B_ * K_);
     const T* Ts = Tg + (size_t)s * B_;
     // Loads are raw (address-clamped, no value select): out-of-range
@@ -1441,7 +1436,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       for (int kk = 0; kk < KS; ++kk) {
         const int k = 4 * kk + q;
         const int kc = (k < K_) ? k : K_ - 1;
-        R.fA[tm][kk] = ldf(&Xs[mc * K_ + kc]);
+        cfA[tm][kk] = ldf(&Xs[mc * K_ + kc]);
       }
     }
 #pragma unroll
@@ -1452,7 +1447,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       for (int ii = 0; ii < BS; ++ii) {
         const int i = 4 * ii + q;
         const int ic = (i < B_) ? i : B_ - 1;
-        R.bB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
+        cbB[tk][ii] = ldf(&Xs[ic * K_ + kc]);
       }
     }
 #pragma unroll
@@ -1460,17 +1455,13 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
 #pragma unroll
       for (int i = 0; i < 4; ++i) {
         const int row = tm * 16 + q * 4 + i;
-        R.tR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
+        ctR[tm][i] = ldf(&Ts[(row < B_) ? row : B_ - 1]);
       }
   };
 
-  prefetch(setA, 0);
-  if (NSETS == 2 && S > 1) prefetch(setB, 1);
+  prefetch(0);
   float loss_last = 0.f;
-  auto body = [&](FSet& R, int s) {
-    const auto& cfA = R.fA;
-    const auto& cbB = R.bB;
-    const auto& ctR = R.tR;
+  for (int s = 0; s < S; ++s) {
     // batch the w reads (LDS; one unconditional read + select per kk)
     float wv[KS];
 #pragma unroll
@@ -1536,10 +1527,10 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
                                                         gacc[tk], 0, 0, 0);
       }
 
-    // this set's regs are dead from here: load step s+NSETS straight into
-    // them — the vmcnt wait attaches to their next use, NSETS-1 full
-    // iterations later
-    if (s + NSETS < S) prefetch(R, s + NSETS);
+    // cur regs are dead from here: load next step's operands straight into
+    // them — the vmcnt wait attaches to their first use (next iteration's
+    // forward MFMA), shadowed by the update/barrier below
+    if (s + 1 < S) prefetch(s + 1);
 
     if (use_mse && loss_out && s == S - 1) loss_last = wave_sum(loss_acc);
 
@@ -1551,7 +1542,7 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
                              mworld, minv_world,
                              seq0 + (unsigned long long)s)) {
         if (lane == 0) *mesh_err = 1u;
-        return false;
+        return;
       }
 #pragma unroll
       for (int tk = 0; tk < KT; ++tk) gacc[tk][0] = gval[tk];
@@ -1570,20 +1561,6 @@ k_toy_multistep_spec(const T* __restrict__ X, const T* __restrict__ Tg,
       }
     }
     __syncthreads();  // ws update visible before next iteration's forward
-    return true;
-  };
-
-  int s = 0;
-  if (NSETS == 2) {
-    for (; s + 2 <= S; s += 2) {
-      if (!body(setA, s)) return;
-      if (!body(setB, s + 1)) return;
-    }
-    if (s < S && !body(setA, s)) return;
-  } else {
-    for (; s < S; ++s) {
-      if (!body(setA, s)) return;
-    }
   }
 
   if (lane < K_) stf(&param[w_off + lane], ws[lane]);
