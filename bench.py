@@ -81,6 +81,7 @@ class DeviceData:
         self._side = torch.cuda.Stream() if (self._cuda and use_side) else None
         self._pending = {}  # epoch -> (xs, ts, ready_event)
         self.bound_epoch = -1  # used by the shard-bound engine path
+        self.bound_block = -1  # used by the epoch-block engine path
         self._fused_shard = os.environ.get("MI355X_SHUFFLE", "") != "randperm"
 
     def _gather(self, epoch):
@@ -123,6 +124,19 @@ class DeviceData:
             ts.record_stream(main)
         self._prep(epoch + 1)
         return xs, ts
+
+    def shards_for_block(self, e0, epochs):
+        """One contiguous buffer pair covering epochs [e0, e0+epochs):
+        block e is exactly shard_for(e0+e)'s content (bitwise). Fused
+        path: ONE gather kernel for the whole block (epoch_shard_multi) —
+        this is what lets the persistent engine's deferral span epoch
+        boundaries (one multistep launch per block instead of per epoch)."""
+        if self._cuda and self._fused_shard:
+            from mi355x_ddp import ops
+            return ops.ext().epoch_shard_multi(self.X, self.T, 1000 + e0,
+                                               epochs, self.rank, self.world)
+        xs, ts = zip(*(self._gather(e0 + e) for e in range(epochs)))
+        return torch.cat(xs), torch.cat(ts)
 
     def batch_for(self, step):
         epoch, s = divmod(step, self.steps_per_epoch)
@@ -238,8 +252,22 @@ def main():
         eng_bind = getattr(engine_obj, "bind_shard", None)
         eng_step_shard = getattr(engine_obj, "step_shard", None)
         data.bound_epoch = -1
+        data.bound_block = -1
 
     make_engine()
+
+    # Epoch-block binding: gather MI355X_EPOCH_BLOCK epochs' shards in ONE
+    # kernel (epoch_shard_multi) and bind them as one shard, so the
+    # engine's deferral — and therefore the multistep kernel launch —
+    # spans epoch boundaries. Default fills the engine's deferral window
+    # (max_defer steps). Needs whole batches per epoch (block row layout)
+    # and the fused shard path; MI355X_EPOCH_BLOCK=1 restores per-epoch.
+    eb = 1
+    if (eng_bind is not None and use_cuda and spe > 0
+            and data.per_rank == spe * args.batch and data._fused_shard):
+        md = getattr(engine_obj, "max_defer", 1024)
+        eb_env = os.environ.get("MI355X_EPOCH_BLOCK")
+        eb = max(1, int(eb_env) if eb_env else md // spe)
 
     def run_steps(start, n):
         if eng_bind is None:
@@ -247,15 +275,26 @@ def main():
                 x, t = data.batch_for(s)
                 step_fn(x, t)
             return
-        cur = data.bound_epoch
         step_shard = eng_step_shard
+        if eb == 1:
+            cur = data.bound_epoch
+            for s in range(start, start + n):
+                e, i = divmod(s, spe)
+                if e != cur:
+                    xs, ts = data.shard_for(e)
+                    eng_bind(xs, ts, args.batch)
+                    data.bound_epoch = cur = e
+                step_shard(i)
+            return
+        cur = data.bound_block
         for s in range(start, start + n):
             e, i = divmod(s, spe)
-            if e != cur:
-                xs, ts = data.shard_for(e)
+            b = e // eb
+            if b != cur:
+                xs, ts = data.shards_for_block(b * eb, eb)
                 eng_bind(xs, ts, args.batch)
-                data.bound_epoch = cur = e
-            step_shard(i)
+                data.bound_block = cur = b
+            step_shard((e - b * eb) * spe + i)
 
     def comm_check():
         if comm is not None and hasattr(comm, "check"):
@@ -361,6 +400,7 @@ def main():
                             "ToyFusedStep": "fused"}[
                                type(engine_obj).__name__]),
                 "loss": args.loss,
+                "epoch_block": eb,
                 "p50_step_ms": p50_ms,
             },
         }

@@ -43,6 +43,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("lr"), py::arg("batch"), py::arg("mesh"));
   m.def("epoch_shard", &mi355x::epoch_shard, py::arg("X"), py::arg("T"),
         py::arg("seed"), py::arg("rank"), py::arg("world"));
+  m.def("epoch_shard_multi", &mi355x::epoch_shard_multi, py::arg("X"),
+        py::arg("T"), py::arg("seed0"), py::arg("epochs"), py::arg("rank"),
+        py::arg("world"));
   m.def("toy_multistep", &mi355x::toy_multistep, py::arg("x"), py::arg("t"),
         py::arg("param_flat"), py::arg("loss_out"), py::arg("use_mse") = true,
         py::arg("w_off") = 0, py::arg("b_off") = 0, py::arg("lr") = 0.0,

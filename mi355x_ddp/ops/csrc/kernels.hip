@@ -808,6 +808,62 @@ std::vector<torch::Tensor> epoch_shard(torch::Tensor X, torch::Tensor Tg,
   return {xs, ts};
 }
 
+// Multi-EPOCH variant: one launch gathers `epochs` consecutive epoch
+// shards (seeds seed0, seed0+1, …) into one contiguous buffer pair —
+// block e of the output is bitwise-identical to epoch_shard(X, T,
+// seed0+e)'s result. This is what lets the persistent-engine deferral
+// span epoch boundaries (one multistep launch per ~max_defer steps
+// instead of one per epoch) and, as a side effect, gives this gather
+// kernel epochs× more workgroups to fill the 256 CUs with.
+template <typename T>
+__global__ void k_epoch_shard_multi(const T* __restrict__ X,
+                                    const T* __restrict__ Tg,
+                                    T* __restrict__ xs, T* __restrict__ ts,
+                                    int n, int K, int per_rank, int rank,
+                                    int world, uint32_t k_mask,
+                                    uint32_t seed0, int epochs) {
+  const long long tid = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  const int cols = K + 1;
+  const long long per_epoch = (long long)per_rank * cols;
+  if (tid >= per_epoch * epochs) return;
+  const int e = (int)(tid / per_epoch);
+  const int t = (int)(tid - (long long)e * per_epoch);
+  const int i = t / cols;
+  const int c = t - i * cols;
+  uint32_t c0, m0, c1, m1;
+  shard_consts(seed0 + (uint32_t)e, &c0, &m0, &c1, &m1);
+  const uint32_t p = (uint32_t)(rank + (size_t)i * world);
+  const uint32_t src = mix_bijection(p, k_mask, (uint32_t)n, c0, m0, c1, m1);
+  const size_t o = (size_t)e * per_rank + i;
+  if (c < K) xs[o * K + c] = X[(size_t)src * K + c];
+  else       ts[o] = Tg[src];
+}
+
+std::vector<torch::Tensor> epoch_shard_multi(torch::Tensor X, torch::Tensor Tg,
+                                             int64_t seed0, int64_t epochs,
+                                             int64_t rank, int64_t world) {
+  TORCH_CHECK(X.is_cuda() && X.is_contiguous() && Tg.is_contiguous());
+  TORCH_CHECK(epochs >= 1 && epochs <= 1 << 20, "epochs out of range");
+  const int n = (int)X.size(0), K = (int)X.size(1);
+  TORCH_CHECK(Tg.size(0) == n && Tg.size(1) == 1, "targets must be [n,1]");
+  const int per_rank = n / (int)world;
+  auto xs = at::empty({epochs * per_rank, K}, X.options());
+  auto ts = at::empty({epochs * per_rank, 1}, Tg.options());
+  uint32_t k_mask = 1;
+  while ((int64_t)k_mask + 1 < n) k_mask = (k_mask << 1) | 1u;
+  const long long total = (long long)epochs * per_rank * (K + 1);
+  DISPATCH_F32_BF16(X.scalar_type(), "epoch_shard_multi", {
+    hipLaunchKernelGGL((k_epoch_shard_multi<scalar_t>),
+                       dim3((uint32_t)((total + 255) / 256)), dim3(256), 0,
+                       cur_stream(), cdptr<scalar_t>(X), cdptr<scalar_t>(Tg),
+                       dptr<scalar_t>(xs), dptr<scalar_t>(ts), n, K, per_rank,
+                       (int)rank, (int)world, k_mask, (uint32_t)seed0,
+                       (int)epochs);
+  });
+  HIP_OK(hipGetLastError());
+  return {xs, ts};
+}
+
 // ---------------------------------------------------------------------------
 // Fused toy training step (fwd + loss-grad + bwd [+ SGD] in ONE kernel):
 // the reference hot loop single_gpu.py:21-26 for model = Linear(K,1).

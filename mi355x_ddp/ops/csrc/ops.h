@@ -70,6 +70,12 @@ std::vector<torch::Tensor> epoch_shard(torch::Tensor X, torch::Tensor Tg,
                                        int64_t seed, int64_t rank,
                                        int64_t world);
 
+// Multi-epoch form: gathers `epochs` consecutive shards (seeds seed0,
+// seed0+1, …) in ONE launch; block e is bitwise epoch_shard(seed0+e).
+std::vector<torch::Tensor> epoch_shard_multi(torch::Tensor X, torch::Tensor Tg,
+                                             int64_t seed0, int64_t epochs,
+                                             int64_t rank, int64_t world);
+
 // Multi-step persistent toy trainer (world-1): x is [S*batch, K] of S
 // consecutive batches; runs S full fwd+loss+bwd+SGD steps in ONE launch
 // with the weights resident in LDS (bitwise-identical per-step arithmetic

@@ -40,3 +40,23 @@ def test_bench_defaults_finish_quickly():
     assert out.returncode == 0, out.stderr[-2000:]
     r = json.loads(out.stdout.strip().splitlines()[-1])
     assert r["steps"] == 2000
+
+
+def test_shards_for_block_matches_per_epoch_concat():
+    # CPU fallback of the multi-epoch gather: block e of shards_for_block
+    # must equal _gather(e0+e) exactly (the GPU kernel path is held to the
+    # same contract bitwise in tests/test_epoch_block_gpu.py)
+    import torch
+
+    sys.path.insert(0, ROOT)
+    import bench
+
+    d = bench.DeviceData(256, rank=1, world=2, batch=8,
+                         device=torch.device("cpu"))
+    xs_b, ts_b = d.shards_for_block(3, 4)
+    per = d.per_rank
+    assert xs_b.shape == (4 * per, 20) and ts_b.shape == (4 * per, 1)
+    for e in range(4):
+        xs, ts = d._gather(3 + e)
+        assert torch.equal(xs_b[e * per:(e + 1) * per], xs)
+        assert torch.equal(ts_b[e * per:(e + 1) * per], ts)
