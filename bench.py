@@ -160,6 +160,37 @@ class DeviceData:
         return self._views[s]
 
 
+def drive_shard_bound(data, spe, batch, eb, bind, step_shard, start, n):
+    """Drive a shard-bound engine for global steps [start, start+n):
+    (re)bind the covering shard — one epoch, or an eb-epoch block — when
+    stepping crosses into a new one, then step by batch INDEX within the
+    bound buffer. Pure index logic, unit-tested on CPU with a recording
+    fake engine (tests/test_bench_contract.py) because the driver's
+    8-GPU scale run exercises shapes (spe=8, eb=128) that never run in
+    the 1-GPU harness. Mid-run starts (the timed region begins at
+    step=warmup, rarely a block boundary) re-enter the current partial
+    block/epoch via the engine's non-sequential step_shard restart."""
+    if eb == 1:
+        cur = data.bound_epoch
+        for s in range(start, start + n):
+            e, i = divmod(s, spe)
+            if e != cur:
+                xs, ts = data.shard_for(e)
+                bind(xs, ts, batch)
+                data.bound_epoch = cur = e
+            step_shard(i)
+        return
+    cur = data.bound_block
+    for s in range(start, start + n):
+        e, i = divmod(s, spe)
+        b = e // eb
+        if b != cur:
+            xs, ts = data.shards_for_block(b * eb, eb)
+            bind(xs, ts, batch)
+            data.bound_block = cur = b
+        step_shard((e - b * eb) * spe + i)
+
+
 def build_engine(kind, comm, lr, device, dtype=torch.float32, use_mse=True):
     from mi355x_ddp.engine import (GraphedToyStep, PersistentToyStep,
                                    ToyFusedStep)
@@ -275,26 +306,8 @@ def main():
                 x, t = data.batch_for(s)
                 step_fn(x, t)
             return
-        step_shard = eng_step_shard
-        if eb == 1:
-            cur = data.bound_epoch
-            for s in range(start, start + n):
-                e, i = divmod(s, spe)
-                if e != cur:
-                    xs, ts = data.shard_for(e)
-                    eng_bind(xs, ts, args.batch)
-                    data.bound_epoch = cur = e
-                step_shard(i)
-            return
-        cur = data.bound_block
-        for s in range(start, start + n):
-            e, i = divmod(s, spe)
-            b = e // eb
-            if b != cur:
-                xs, ts = data.shards_for_block(b * eb, eb)
-                eng_bind(xs, ts, args.batch)
-                data.bound_block = cur = b
-            step_shard((e - b * eb) * spe + i)
+        drive_shard_bound(data, spe, args.batch, eb, eng_bind,
+                          eng_step_shard, start, n)
 
     def comm_check():
         if comm is not None and hasattr(comm, "check"):
