@@ -54,7 +54,7 @@ def _env_mb(name: str, default: float) -> float:
 class Bucket:
     __slots__ = ("index", "params", "offsets", "numel", "flat_param",
                  "flat_grad", "pending", "ready", "plan", "nblocks",
-                 "plan_grads")
+                 "plan_grads", "plan_params")
 
     def __init__(self, index: int, params: List[torch.nn.Parameter],
                  offsets: List[int], numel: int, device: torch.device,
@@ -70,6 +70,7 @@ class Bucket:
         self.plan = None      # device copy plan for the flatten kernel
         self.nblocks = 0
         self.plan_grads = None  # keeps the planned grad tensors alive
+        self.plan_params = None  # members included in the plan, in order
 
     def grad_view(self, i: int) -> torch.Tensor:
         p = self.params[i]
@@ -147,7 +148,6 @@ class Reducer:
             for b in self.buckets for p in b.params
         ]
         self._next_launch = 0
-        self._in_backward = False
 
     # -- wrap-time state sync --------------------------------
     def broadcast_params(self, root: int = 0) -> None:
@@ -158,11 +158,8 @@ class Reducer:
 
     # -- backward-side machinery ----------------------------------------
     def _make_hook(self, p: torch.nn.Parameter):
-        bucket, i = None, None  # resolved lazily to avoid capturing big closures
-
         def hook(param: torch.nn.Parameter) -> None:
             b, idx = self._param_index[param]
-            self._in_backward = True
             if self.grad_views:
                 g = param.grad
                 view = b.grad_view(idx)
@@ -191,18 +188,32 @@ class Reducer:
 
     def _flatten_bucket(self, b: Bucket) -> None:
         if b.flat_grad.is_cuda:
+            # The copy plan captures raw grad POINTERS, so it is only valid
+            # while every member grad is still the same tensor autograd
+            # accumulates into. If any was replaced (p.grad = None between
+            # steps, out-of-place accumulation) or a previously-unused
+            # member now has a grad, rebuild — a stale plan would silently
+            # flatten last iteration's memory or drop the new gradient.
+            if b.plan is not None and (
+                    any(p.grad is not g
+                        for p, g in zip(b.plan_params, b.plan_grads))
+                    or sum(p.grad is not None for p in b.params)
+                    != len(b.plan_params)):
+                b.plan = None
             if b.plan is None:
-                grads, offs = [], []
+                grads, offs, members = [], [], []
                 for i, p in enumerate(b.params):
                     if p.grad is None:
                         continue  # unused param: its segment stays zero
                     assert p.grad.is_contiguous(), "grad must be contiguous"
                     grads.append(p.grad)
                     offs.append(b.offsets[i])
+                    members.append(p)
                 b.plan = ops.ext().build_copy_plan(
                     grads, offs, b.flat_grad.device.index or 0)
                 b.nblocks = b.plan.shape[0]
                 b.plan_grads = grads  # plan holds raw pointers: keep alive
+                b.plan_params = members
             # one launch: gather member grads into the bucket AND zero the
             # sources (the reference's implicit flatten + zero_grad)
             ops.ext().flatten_into(b.flat_grad, b.plan, b.nblocks, True)
@@ -228,7 +239,6 @@ class Reducer:
             b.pending = len(b.params)
             b.ready = False
         self._next_launch = 0
-        self._in_backward = False
 
     # -- introspection ----------------------------------------------------
     def flat_pairs(self):

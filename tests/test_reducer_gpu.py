@@ -120,3 +120,43 @@ def test_ddp_multibucket_world2_on_device(grad_views, tmp_path):
     for p, pr in zip(got, ref.parameters()):
         assert torch.allclose(p, pr.detach().cpu(), atol=1e-5, rtol=1e-4), \
             (p - pr.detach().cpu()).abs().max()
+
+
+def test_copy_mode_plan_rebuilds_after_grad_replacement():
+    """Copy mode captures raw grad pointers in its flatten plan; if the
+    user replaces a grad tensor between steps (p.grad = None forces
+    autograd to allocate a fresh one), the plan must be rebuilt — a stale
+    plan would silently flatten last iteration's memory."""
+    x = torch.randn(16, 64, device=DEV)
+    t = torch.randn(16, 10, device=DEV)
+
+    ref = _mlp(3)
+    opt_ref = torch.optim.SGD(ref.parameters(), lr=0.01)
+    for step in range(4):
+        opt_ref.zero_grad(set_to_none=True)
+        torch.nn.functional.mse_loss(ref(x), t).backward()
+        opt_ref.step()
+
+    model = _mlp(3)
+    red = Reducer(list(model.parameters()), comm=None,
+                  bucket_cap_mb=0.5, grad_views=False)
+    opt = FusedSGD(model.parameters(), lr=0.01)
+    opt.attach_reducer(red)
+    plans_before = None
+    for step in range(4):
+        if step == 2:
+            # force autograd to allocate fresh grad tensors mid-run
+            for p in model.parameters():
+                p.grad = None
+            plans_before = [b.plan for b in red.buckets]
+        torch.nn.functional.mse_loss(model(x), t).backward()
+        red.finalize()
+        opt.step()
+    torch.cuda.synchronize()
+    # every plan was rebuilt after the replacement...
+    assert all(b.plan is not old
+               for b, old in zip(red.buckets, plans_before))
+    # ...and the training history matches plain autograd
+    for p, pr in zip(model.parameters(), ref.parameters()):
+        assert torch.allclose(p, pr, atol=1e-5, rtol=1e-4), \
+            (p.shape, (p - pr).abs().max())
