@@ -32,12 +32,16 @@ def ddp_setup(rank: Optional[int] = None, world_size: Optional[int] = None,
       sets MASTER_ADDR/PORT defaults then init_process_group)
     - env-var style under torchrun (reference multigpu_torchrun.py:12-13).
 
-    The backend defaults to "nccl" on GPU hosts (the string is kept for the
-    torchrun contract — tensor traffic still goes through RcclComm) and
-    "gloo" on CPU-only hosts.
+    The backend defaults to the compound map "cpu:gloo,cuda:nccl" on GPU
+    hosts — GPU tensors ride the c10d NCCL(=RCCL) backend exactly as the
+    reference's "nccl" does (gradient bytes still go through RcclComm),
+    while CPU tensors (the transport-ladder agreement flags and the mesh's
+    gloo cross-validation, see all_ranks_agree/P2pMeshComm.validate) ride
+    gloo. A bare "nccl" group would reject those CPU collectives. CPU-only
+    hosts use "gloo".
     """
     if backend is None:
-        backend = "nccl" if torch.cuda.is_available() else "gloo"
+        backend = "cpu:gloo,cuda:nccl" if torch.cuda.is_available() else "gloo"
     if rank is not None and world_size is not None:
         os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
         os.environ.setdefault("MASTER_PORT", "12355")
@@ -164,12 +168,18 @@ class P2pMeshComm:
             self.validate()
 
     def validate(self) -> None:
-        """One mesh all-reduce cross-checked against the gloo group."""
+        """One mesh all-reduce cross-checked against the c10d group (gloo
+        normally; the c10d NCCL backend when the default group is
+        nccl-only — still a transport independent of this mesh)."""
         probe = torch.arange(24, dtype=torch.float32, device=self._device)
         probe = probe * (self.rank + 1)
         ref = probe.cpu()
-        dist.all_reduce(ref, op=dist.ReduceOp.SUM)
-        ref = ref / self.world
+        try:
+            dist.all_reduce(ref, op=dist.ReduceOp.SUM)
+        except RuntimeError:
+            ref = probe.clone()  # device tensor for an nccl-only group
+            dist.all_reduce(ref, op=dist.ReduceOp.SUM)
+        ref = ref.cpu() / self.world
         self._mesh.all_reduce_avg_inline(probe)
         torch.cuda.synchronize()
         self._mesh.check()
@@ -210,7 +220,14 @@ def all_ranks_agree(ok: bool) -> bool:
     transport decision must therefore be agreed by all ranks — call this
     at the same program point on every rank."""
     t = torch.tensor([1 if ok else 0])
-    dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    try:
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+    except RuntimeError:
+        # nccl-only default group (user-initialized): agree via a device
+        # tensor instead. Deterministic across ranks — every rank sees the
+        # same group backend, so every rank takes the same branch.
+        t = t.cuda()
+        dist.all_reduce(t, op=dist.ReduceOp.MIN)
     return bool(t.item())
 
 

@@ -150,3 +150,27 @@ def test_reducer_bucket_assignment():
     # grads were zeroed by the fused step
     for fp, fg in r.flat_pairs():
         assert fg.abs().sum() == 0
+
+
+def test_ddp_setup_backend_defaults(monkeypatch):
+    """GPU hosts must get the compound backend map: a bare "nccl" group
+    rejects the CPU tensors that all_ranks_agree / mesh validation
+    all-reduce over the default group (latent crash for every world>1
+    entrypoint run on a real multi-GPU node)."""
+    from mi355x_ddp.parallel import comm as C
+
+    calls = {}
+    monkeypatch.setenv("MASTER_ADDR", "127.0.0.1")
+    monkeypatch.setenv("MASTER_PORT", "29999")
+    monkeypatch.setattr(
+        C.dist, "init_process_group",
+        lambda backend, **kw: calls.__setitem__("backend", backend))
+
+    monkeypatch.setattr(C.torch.cuda, "is_available", lambda: False)
+    C.ddp_setup(0, 1)
+    assert calls["backend"] == "gloo"
+
+    monkeypatch.setattr(C.torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(C.torch.cuda, "set_device", lambda *_: None)
+    C.ddp_setup(0, 1)
+    assert calls["backend"] == "cpu:gloo,cuda:nccl"
