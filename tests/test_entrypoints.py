@@ -50,6 +50,21 @@ def test_multigpu_spawn_script(tmp_path):
     assert (tmp_path / "checkpoint.pt").exists()
 
 
+def test_multigpu_spawn_bf16_variant(tmp_path):
+    # BASELINE.json config 2 names the bf16 variant of this entrypoint;
+    # MI355X_DTYPE=bf16 casts the model (HipLinear re-casts the f32 loader
+    # batches, the losses re-cast targets) and the checkpoint stays loadable
+    r = _run([sys.executable, os.path.join(ROOT, "multigpu.py"), "1", "1"],
+             cwd=tmp_path, env_extra={"MI355X_WORLD": "2",
+                                      "MI355X_DTYPE": "bf16",
+                                      "MASTER_PORT": _free_port()})
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "[GPU0] Epoch 0" in r.stdout and "[GPU1] Epoch 0" in r.stdout
+    import torch
+    sd = torch.load(str(tmp_path / "checkpoint.pt"), weights_only=True)
+    assert sd["weight"].dtype == torch.bfloat16
+
+
 def test_torchrun_snapshot_resume(tmp_path):
     script = os.path.join(ROOT, "multigpu_torchrun.py")
     base = TORCHRUN + ["--standalone", "--local-addr", "127.0.0.1",
