@@ -181,3 +181,23 @@ def test_drive_shard_bound_odd_block_vs_defer_window():
     # eb*spe NOT a multiple of max_defer: engine auto-flushes mid-block
     executed, data, eng = _replay(8, 100, [(0, 3000)], max_defer=512)
     assert executed == _expected(8, [(0, 3000)])
+
+
+def test_drive_shard_bound_property_fuzz():
+    # randomized spe/eb/max_defer/segment boundaries: the executed
+    # (epoch, step) stream must always equal the requested one, and no
+    # launch may index past its bound buffer (asserted inside _replay)
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=60, deadline=None)
+    @given(st.integers(1, 96), st.integers(1, 160), st.integers(1, 2048),
+           st.lists(st.integers(1, 700), min_size=1, max_size=4))
+    def run(spe, eb, md, seg_lens):
+        segments, start = [], 0
+        for n in seg_lens:
+            segments.append((start, n))
+            start += n
+        executed, _, _ = _replay(spe, eb, segments, max_defer=md)
+        assert executed == _expected(spe, segments)
+
+    run()
