@@ -181,12 +181,31 @@ class Trainer:
         self.epochs_run = snapshot["EPOCHS_RUN"]
         print(f"Resuming training from snapshot at Epoch {self.epochs_run}")
 
+    @staticmethod
+    def _atomic_save(obj, path: str) -> None:
+        """torch.save via tmp + os.replace. The reference writes the live
+        file in place (multigpu_torchrun.py:61): a worker crash mid-save
+        corrupts snapshot.pt, and every elastic restart then dies loading
+        it — a permanent crash loop. Rename is atomic on POSIX, so readers
+        (including other nodes on a shared mount, SURVEY §2.1 'multinode
+        snapshot race') see either the old or the new complete file."""
+        tmp = f"{path}.tmp.{os.getpid()}"
+        try:
+            torch.save(obj, tmp)
+            os.replace(tmp, path)
+        except BaseException:
+            try:
+                os.remove(tmp)
+            except OSError:
+                pass
+            raise
+
     def _save_snapshot(self, epoch: int) -> None:
         snapshot = {
             "MODEL_STATE": self._unwrapped().state_dict(),
             "EPOCHS_RUN": epoch,
         }
-        torch.save(snapshot, self.snapshot_path)
+        self._atomic_save(snapshot, self.snapshot_path)
         print(f"Epoch {epoch} | Training snapshot saved at {self.snapshot_path}")
 
     def _save_checkpoint(self, epoch: int) -> None:
@@ -194,7 +213,7 @@ class Trainer:
             ckp = self.model.state_dict()  # DDP-prefixed, multigpu_profile.py:76-78
         else:
             ckp = self._unwrapped().state_dict()
-        torch.save(ckp, self.checkpoint_path)
+        self._atomic_save(ckp, self.checkpoint_path)
         print(f"Epoch {epoch} | Training checkpoint saved at {self.checkpoint_path}")
 
     # -- the hot loop -----------------------------------------------------

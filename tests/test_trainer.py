@@ -134,3 +134,32 @@ def test_reference_written_snapshot_loads(tmp_chdir):
     assert tr.epochs_run == 5
     assert torch.equal(model.weight.detach(), ref_model.weight.detach())
     assert torch.equal(model.bias.detach(), ref_model.bias.detach())
+
+
+def test_snapshot_save_is_atomic(tmp_chdir, monkeypatch):
+    """A crash mid-save must leave the previous snapshot intact (the
+    reference writes the live file in place — a mid-save kill corrupts
+    snapshot.pt and turns every elastic restart into a crash loop)."""
+    tr = _make(snapshot_path="snapshot.pt")
+    tr._save_snapshot(3)
+    good = torch.load("snapshot.pt", weights_only=True)
+    assert good["EPOCHS_RUN"] == 3
+
+    real_save = torch.save
+
+    def crashing_save(obj, path):
+        with open(path, "wb") as f:
+            f.write(b"partial garbage")  # simulate dying mid-write
+        raise RuntimeError("killed mid-save")
+
+    monkeypatch.setattr(torch, "save", crashing_save)
+    try:
+        tr._save_snapshot(4)
+    except RuntimeError:
+        pass
+    monkeypatch.setattr(torch, "save", real_save)
+
+    # the live file still holds the epoch-3 snapshot, and no tmp debris
+    again = torch.load("snapshot.pt", weights_only=True)
+    assert again["EPOCHS_RUN"] == 3
+    assert not [f for f in os.listdir(".") if ".tmp." in f]
