@@ -9,6 +9,10 @@ site) is fixed.
 Set MI355X_DTYPE=bf16 for the bf16 variant (BASELINE.json config 2).
 CPU fallback (no GPU): world size from MI355X_WORLD (default 2) over gloo —
 used by the CPU test tier.
+
+MI355X_ENGINE=persistent|fused|auto engages the fast toy engines through this
+entrypoint (silently falls back to the generic hooks path when the
+model/loss/device do not qualify — e.g. on CPU, or for the CE-loss stages).
 """
 
 import os
@@ -41,7 +45,8 @@ def main(rank: int, world_size: int, total_epochs: int, save_every: int):
                                         distributed=True,
                                         num_replicas=world_size, rank=rank)
         device = rank if torch.cuda.is_available() else "cpu"
-        trainer = Trainer(model, train_data, optimizer, device, save_every)
+        trainer = Trainer(model, train_data, optimizer, device, save_every,
+                          engine=os.environ.get("MI355X_ENGINE", "hooks"))
         trainer.train(total_epochs)
     finally:
         destroy_process_group()

@@ -8,8 +8,13 @@ Honors the torchrun env contract (RANK/LOCAL_RANK/WORLD_SIZE/MASTER_*,
 SURVEY §2.2 N14) so the stock elastic agent supervises this engine: on a
 worker failure torchrun restarts everyone, each worker finds snapshot.pt
 and resumes from EPOCHS_RUN (reference flow).
+
+MI355X_ENGINE=persistent|fused|auto engages the fast toy engines through this
+entrypoint (silently falls back to the generic hooks path when the
+model/loss/device do not qualify — e.g. on CPU, or for the CE-loss stages).
 """
 
+import os
 import sys
 
 import torch
@@ -35,7 +40,8 @@ def main(total_epochs: int, save_every: int, snapshot_path: str = "snapshot.pt")
         train_data = prepare_dataloader(dataset, batch_size=32, distributed=True)
         device = None if torch.cuda.is_available() else "cpu"  # None -> LOCAL_RANK
         trainer = Trainer(model, train_data, optimizer, device, save_every,
-                          snapshot_path=snapshot_path)
+                          snapshot_path=snapshot_path,
+                          engine=os.environ.get("MI355X_ENGINE", "hooks"))
         trainer.train(total_epochs)
     finally:
         destroy_process_group()

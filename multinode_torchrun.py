@@ -8,8 +8,13 @@ Differences from stage 3, kept for parity: MSE loss
 (reference multinode_torchrun.py:46) and the banner shows the GLOBAL rank
 (reference :25,52). The reference's snapshot race (every node's local rank
 0 writing the shared snapshot, :68) is fixed: only global rank 0 saves.
+
+MI355X_ENGINE=persistent|fused|auto engages the fast toy engines through this
+entrypoint (silently falls back to the generic hooks path when the
+model/loss/device do not qualify — e.g. on CPU, or for the CE-loss stages).
 """
 
+import os
 import sys
 
 import torch
@@ -35,7 +40,8 @@ def main(total_epochs: int, save_every: int, snapshot_path: str = "snapshot.pt")
         train_data = prepare_dataloader(dataset, batch_size=32, distributed=True)
         device = None if torch.cuda.is_available() else "cpu"
         trainer = Trainer(model, train_data, optimizer, device, save_every,
-                          snapshot_path=snapshot_path, loss_fn="mse")
+                          snapshot_path=snapshot_path, loss_fn="mse",
+                          engine=os.environ.get("MI355X_ENGINE", "hooks"))
         trainer.train(total_epochs)
     finally:
         destroy_process_group()

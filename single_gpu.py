@@ -3,8 +3,13 @@
 Same CLI (`python single_gpu.py <total_epochs> <save_every>`), same banner,
 same `checkpoint.pt` raw-state_dict format. Runs on GPU 0 when a GPU is
 present, else on CPU (BASELINE.json config 1: the CPU plumbing path).
+
+MI355X_ENGINE=persistent|fused|auto engages the fast toy engines through this
+entrypoint (silently falls back to the generic hooks path when the
+model/loss/device do not qualify — e.g. on CPU, or for the CE-loss stages).
 """
 
+import os
 import sys
 
 import torch
@@ -27,7 +32,8 @@ def main(device, total_epochs: int, save_every: int):
     dataset, model, optimizer = load_train_objs()
     train_data = prepare_dataloader(dataset, batch_size=32)
     trainer = Trainer(model, train_data, optimizer, device, save_every,
-                      wrap_ddp=False)
+                      wrap_ddp=False,
+                      engine=os.environ.get("MI355X_ENGINE", "hooks"))
     trainer.train(total_epochs)
 
 
