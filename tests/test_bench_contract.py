@@ -201,3 +201,24 @@ def test_drive_shard_bound_property_fuzz():
         assert executed == _expected(spe, segments)
 
     run()
+
+
+def test_bench_world2_gloo_driver_form():
+    # the driver's multi-GPU launch form, on a 2-rank CPU/gloo world:
+    # rank 0 must print ONE whole-job JSON line (value aggregated over
+    # ranks, elapsed = max over ranks)
+    port = 29500 + os.getpid() % 500
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", str(port), os.path.join(ROOT, "bench.py"),
+         "--gpus", "2", "--steps", "20", "--warmup", "4",
+         "--p50-probes", "2"],
+        cwd=ROOT, capture_output=True, text=True, timeout=420)
+    assert out.returncode == 0, out.stderr[-2000:]
+    lines = [l for l in out.stdout.strip().splitlines() if l.startswith("{")]
+    assert len(lines) == 1, out.stdout[-1000:]
+    r = json.loads(lines[0])
+    assert r["config"]["parallelism"] == "dp2"
+    assert r["config"]["comm"] == "gloo-cpu"
+    assert r["value"] > 0
