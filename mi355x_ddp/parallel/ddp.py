@@ -17,6 +17,8 @@ single-GPU benchmark runs the same engine.
 
 from __future__ import annotations
 
+from contextlib import contextmanager
+
 from typing import Optional
 
 import torch
@@ -59,6 +61,21 @@ class DDP(torch.nn.Module):
         """Launch straggler buckets and fence compute on the comm stream.
         Must be called between loss.backward() and optimizer.step()."""
         self.reducer.finalize()
+
+    @contextmanager
+    def no_sync(self):
+        """Gradient accumulation, stock-DDP style (beyond reference parity):
+        backward passes inside the context skip the bucket all-reduces;
+        gradients accumulate locally (flat-bucket views, or p.grad in copy
+        mode) and the first backward OUTSIDE the context reduces the
+        accumulated totals. Do not call optimizer.step() inside the
+        context, and call finalize_backward() after every backward (it
+        resets the per-step ready counters either way)."""
+        self.reducer.skip_comm = True
+        try:
+            yield
+        finally:
+            self.reducer.skip_comm = False
 
     # parity helper: stock DDP state_dict carries the "module." prefix;
     # nn.Module gives us that for free since `module` is a submodule.

@@ -148,6 +148,7 @@ class Reducer:
             for b in self.buckets for p in b.params
         ]
         self._next_launch = 0
+        self.skip_comm = False  # set by DDP.no_sync() during accumulation
 
     # -- wrap-time state sync --------------------------------
     def broadcast_params(self, root: int = 0) -> None:
@@ -180,9 +181,11 @@ class Reducer:
         while (self._next_launch < len(self.buckets)
                and self.buckets[self._next_launch].ready):
             b = self.buckets[self._next_launch]
-            if not self.grad_views:
+            if not self.grad_views and not self.skip_comm:
+                # during no_sync, leave grads accumulating in p.grad —
+                # flatten zeroes its sources, which would drop them
                 self._flatten_bucket(b)
-            if self.comm is not None:
+            if self.comm is not None and not self.skip_comm:
                 self.comm.all_reduce_avg(b.flat_grad)
             self._next_launch += 1
 
@@ -228,12 +231,12 @@ class Reducer:
         """Called after loss.backward(): launch stragglers, fence compute on
         the comm stream, and reset per-step state."""
         for b in self.buckets[self._next_launch:]:
-            if not self.grad_views:
+            if not self.grad_views and not self.skip_comm:
                 self._flatten_bucket(b)
-            if self.comm is not None:
+            if self.comm is not None and not self.skip_comm:
                 self.comm.all_reduce_avg(b.flat_grad)
         self._next_launch = len(self.buckets)
-        if self.comm is not None:
+        if self.comm is not None and not self.skip_comm:
             self.comm.join_compute()
         for b in self.buckets:
             b.pending = len(b.params)

@@ -174,3 +174,77 @@ def test_ddp_setup_backend_defaults(monkeypatch):
     monkeypatch.setattr(C.torch.cuda, "set_device", lambda *_: None)
     C.ddp_setup(0, 1)
     assert calls["backend"] == "cpu:gloo,cuda:nccl"
+
+
+def _accum_worker(rank, seed, data, port, grad_views, out_path):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        model = _make_model(seed)
+        engine = DDP(model, grad_views=grad_views)
+        opt = FusedSGD(model.parameters(), lr=LR)
+        opt.attach_reducer(engine.reducer)
+        half = 16
+        # pairs of micro-batches: first under no_sync (local accumulation),
+        # second with communication (reduces the accumulated totals)
+        for (x1, t1), (x2, t2) in zip(data[0::2], data[1::2]):
+            with engine.no_sync():
+                loss = ops.mse_loss(
+                    engine(x1[rank * half:(rank + 1) * half]),
+                    t1[rank * half:(rank + 1) * half])
+                loss.backward()
+                engine.finalize_backward()
+            loss = ops.mse_loss(
+                engine(x2[rank * half:(rank + 1) * half]),
+                t2[rank * half:(rank + 1) * half])
+            loss.backward()
+            engine.finalize_backward()
+            opt.step()
+        if rank == 0:
+            torch.save([p.detach().clone() for p in model.parameters()],
+                       out_path)
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.parametrize("grad_views", [True, False])
+def test_no_sync_gradient_accumulation(grad_views, tmp_path):
+    """DDP.no_sync accumulation == single-process training whose gradient
+    is mean-over-shards of (g(micro1) + g(micro2))."""
+    seed = 11
+    data = _make_data(seed)  # 5 steps -> 2 accumulation pairs
+    # single-process reference computing the identical update
+    model = _make_model(seed)
+    opt = torch.optim.SGD(model.parameters(), lr=LR)
+    half = 16
+    for (x1, t1), (x2, t2) in zip(data[0::2], data[1::2]):
+        opt.zero_grad()
+        acc = None
+        for x, t in ((x1, t1), (x2, t2)):
+            for r in range(WORLD):
+                xs = x[r * half:(r + 1) * half]
+                ts = t[r * half:(r + 1) * half]
+                gs = torch.autograd.grad(
+                    ops.mse_loss(model(xs), ts), list(model.parameters()))
+                gs = [g / WORLD for g in gs]
+                acc = gs if acc is None else [a + g for a, g in zip(acc, gs)]
+        for p, g in zip(model.parameters(), acc):
+            p.grad = g
+        opt.step()
+    want = [p.detach().clone() for p in model.parameters()]
+
+    out_path = os.path.join(str(tmp_path), "accum.pt")
+    ctx = mp.get_context("spawn")
+    port = _free_port()
+    procs = [ctx.Process(target=_accum_worker,
+                         args=(r, seed, data, port, grad_views, out_path))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    got = torch.load(out_path, weights_only=True)
+    for g, w in zip(got, want):
+        assert torch.allclose(g, w, atol=1e-6), (g - w).abs().max()
