@@ -248,3 +248,44 @@ def test_no_sync_gradient_accumulation(grad_views, tmp_path):
     got = torch.load(out_path, weights_only=True)
     for g, w in zip(got, want):
         assert torch.allclose(g, w, atol=1e-6), (g - w).abs().max()
+
+
+def _bufsync_worker(rank, port, out_path):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        torch.manual_seed(3)
+        m = torch.nn.Sequential(torch.nn.BatchNorm1d(8), toy_model(8, 1))
+        engine = DDP(m)
+        bn = m[0]
+        if rank == 1:  # drift rank 1's running stats
+            with torch.no_grad():
+                bn.running_mean.add_(5.0)
+                bn.num_batches_tracked.add_(7)
+        engine.eval()  # eval: forward does not update the stats itself
+        engine(torch.rand(4, 8))
+        if rank == 1:
+            torch.save({"mean": bn.running_mean.clone(),
+                        "nbt": bn.num_batches_tracked.clone()}, out_path)
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_buffer_broadcast_every_forward(tmp_path):
+    """stock-DDP broadcast_buffers parity: rank 0's buffers (fp32 running
+    stats AND the int64 num_batches_tracked) overwrite drifted replicas
+    before every forward."""
+    out_path = os.path.join(str(tmp_path), "bufsync.pt")
+    ctx = mp.get_context("spawn")
+    port = _free_port()
+    procs = [ctx.Process(target=_bufsync_worker, args=(r, port, out_path))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    got = torch.load(out_path, weights_only=True)
+    assert torch.all(got["mean"] == 0.0)  # rank 0's pristine running_mean
+    assert int(got["nbt"]) == 0
