@@ -29,10 +29,28 @@ from .reducer import Reducer
 
 class DDP(torch.nn.Module):
     def __init__(self, module: torch.nn.Module,
+                 device_ids=None, output_device=None,
                  bucket_cap_mb: Optional[float] = None,
                  comm=None, grad_views: Optional[bool] = None,
                  broadcast_buffers: bool = True):
+        # device_ids/output_device: accepted for drop-in compatibility with
+        # the stock signature the reference uses (`DDP(model,
+        # device_ids=[gpu_id])`, ref multigpu.py:36). The model must already
+        # live on that device (our Trainer moves it first, as the reference
+        # does); a mismatch is an error, not a silent migration.
         super().__init__()
+        if device_ids:
+            if len(device_ids) != 1:
+                raise ValueError("single-device module expected (like stock "
+                                 "DDP's single-process-single-device mode)")
+            want = torch.device("cuda", device_ids[0]) \
+                if isinstance(device_ids[0], int) else torch.device(device_ids[0])
+            for p in module.parameters():
+                if p.device != want:
+                    raise ValueError(
+                        f"model is on {p.device} but device_ids names {want}; "
+                        "move the model first (model.to(device))")
+                break
         self.module = module
         params = list(module.parameters())
         device = params[0].device if params else torch.device("cpu")

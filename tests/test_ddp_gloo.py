@@ -289,3 +289,76 @@ def test_buffer_broadcast_every_forward(tmp_path):
     got = torch.load(out_path, weights_only=True)
     assert torch.all(got["mean"] == 0.0)  # rank 0's pristine running_mean
     assert int(got["nbt"]) == 0
+
+
+def test_ddp_accepts_stock_signature():
+    """The reference wraps with `DDP(model, device_ids=[gpu_id])`
+    (ref multigpu.py:36); drop-in users keep that call form. On CPU,
+    device_ids=["cpu"] validates; a device mismatch is loud."""
+    m = toy_model(20, 1)
+    DDP(m, device_ids=["cpu"])  # validates, no-op
+    m2 = toy_model(20, 1)
+    with pytest.raises(ValueError):
+        DDP(m2, device_ids=[0])  # names cuda:0, model is on CPU
+
+
+class _MaybeUnused(torch.nn.Module):
+    def __init__(self):
+        super().__init__()
+        self.a = toy_model(20, 4)
+        self.b = toy_model(4, 1)
+        self.extra = toy_model(20, 1)  # unused on even steps
+        self.use_extra = False
+
+    def forward(self, x):
+        y = self.b(torch.relu(self.a(x)))
+        if self.use_extra:
+            y = y + self.extra(x)
+        return y
+
+
+def _unused_worker(rank, port, out_path):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        torch.manual_seed(5)
+        m = _MaybeUnused()
+        engine = DDP(m)
+        opt = FusedSGD(m.parameters(), lr=LR)
+        opt.attach_reducer(engine.reducer)
+        g = torch.Generator().manual_seed(50)
+        for step in range(4):
+            m.use_extra = step % 2 == 1  # extra's params unused half the time
+            x = torch.rand(32, 20, generator=g)
+            t = torch.rand(32, 1, generator=g)
+            half = 16
+            loss = ops.mse_loss(
+                engine(x[rank * half:(rank + 1) * half]),
+                t[rank * half:(rank + 1) * half])
+            loss.backward()
+            engine.finalize_backward()  # launches straggler buckets
+            opt.step()
+        if rank == 0:
+            torch.save([p.detach().clone() for p in m.parameters()], out_path)
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_unused_parameters_do_not_hang(tmp_path):
+    """A model whose submodule is unused on some steps trains without
+    find_unused_parameters-style flags: finalize() is an explicit
+    all-ranks sync point, so straggler buckets launch in order with zero
+    gradient segments (stock DDP HANGS here unless told in advance)."""
+    out_path = os.path.join(str(tmp_path), "unused.pt")
+    ctx = mp.get_context("spawn")
+    port = _free_port()
+    procs = [ctx.Process(target=_unused_worker, args=(r, port, out_path))
+             for r in range(WORLD)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=180)
+        assert p.exitcode == 0
+    got = torch.load(out_path, weights_only=True)
+    assert all(torch.isfinite(p).all() for p in got)
