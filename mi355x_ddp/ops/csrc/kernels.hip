@@ -21,7 +21,9 @@
 // trainers (`k_toy_multistep_spec`, `k_toy_multistep_bf16w`) run S
 // sequential SGD steps per launch with LDS-resident weights; their MESH
 // variants embed a per-step device-side xGMI all-reduce (p2p_mesh.h);
-// `k_epoch_shard` fuses the epoch shuffle into one copy pass.
+// `k_epoch_shard` fuses the epoch shuffle into one copy pass, and
+// `k_epoch_shard_multi` gathers a whole BLOCK of epochs per launch so
+// the engine's deferral spans epoch boundaries (profiles r01q).
 
 #include <hip/hip_runtime.h>
 #include <hip/hip_bf16.h>
