@@ -82,9 +82,13 @@ class Trainer:
     ) -> None:
         """engine: "hooks" (default — the generic autograd+reducer path,
         reference semantics for arbitrary models), or "fused"/"persistent"
-        /"auto" to run the toy fast path (single-launch fused step /
-        multi-step deferred kernel) when the model+loss+optimizer qualify;
-        silently falls back to hooks otherwise."""
+        /"graph"/"auto" to run the toy fast path (single-launch fused step /
+        multi-step deferred kernel / hipGraph replay) when the
+        model+loss+optimizer qualify; silently falls back to hooks
+        otherwise. Unknown values raise."""
+        if engine not in ("hooks", "auto", "fused", "persistent", "graph"):
+            raise ValueError(f"unknown engine {engine!r} (hooks | auto | "
+                             "fused | persistent | graph)")
         if gpu_id is None:
             gpu_id = int(os.environ.get("LOCAL_RANK", 0))
         self.gpu_id = gpu_id
@@ -157,12 +161,13 @@ class Trainer:
             comm = None
         if kind == "auto":
             kind = "persistent"
-        from .engine import PersistentToyStep, ToyFusedStep
+        from .engine import GraphedToyStep, PersistentToyStep, ToyFusedStep
         lr = self.optimizer.param_groups[0]["lr"]
         if kind == "persistent":
             if comm is not None and getattr(comm, "_mesh", None) is None:
                 kind = "fused"  # multi-step needs world 1 or a mesh comm
-        cls = PersistentToyStep if kind == "persistent" else ToyFusedStep
+        cls = {"persistent": PersistentToyStep, "graph": GraphedToyStep,
+               "fused": ToyFusedStep}[kind]
         self._engine = cls(m, comm=comm, lr=lr, use_mse=True)
         if comm is not None:
             self._engine.reducer.broadcast_params(root=0)

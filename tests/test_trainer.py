@@ -163,3 +163,13 @@ def test_snapshot_save_is_atomic(tmp_chdir, monkeypatch):
     again = torch.load("snapshot.pt", weights_only=True)
     assert again["EPOCHS_RUN"] == 3
     assert not [f for f in os.listdir(".") if ".tmp." in f]
+
+
+def test_unknown_engine_rejected():
+    import pytest
+    ds = ToyDataset(64, seed=0)
+    model = toy_model(20, 1)
+    opt = FusedSGD(model.parameters(), lr=1e-3)
+    dl = prepare_dataloader(ds, 32)
+    with pytest.raises(ValueError, match="unknown engine"):
+        Trainer(model, dl, opt, "cpu", 1, wrap_ddp=False, engine="presistent")
