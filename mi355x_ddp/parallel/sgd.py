@@ -27,7 +27,13 @@ class FusedSGD(torch.optim.Optimizer):
 
     def attach_reducer(self, reducer) -> None:
         """Bind the reducer's flat (param, grad) pairs; step() then runs one
-        fused kernel per bucket instead of one update per parameter."""
+        fused kernel per bucket instead of one update per parameter.
+        Requires a single param_group: the flat-bucket step applies ONE lr
+        to each bucket, and the per-group loop in step() would otherwise
+        re-apply every bucket once per group."""
+        if len(self.param_groups) != 1:
+            raise ValueError("FusedSGD.attach_reducer requires a single "
+                             "param_group (flat buckets take one lr)")
         self._flat_pairs = reducer.flat_pairs()
         self._bucketed = {id(p) for b in reducer.buckets for p in b.params}
 

@@ -105,3 +105,21 @@ def test_transport_agreement_on_partial_failure(tmp_path):
     for r in range(2):
         d = torch.load(tmp_path / f"agree{r}.pt", weights_only=True)
         assert d["kind"] == "gloo-fallback"
+
+
+def test_fused_sgd_rejects_multiple_groups_on_attach():
+    # step() applies each flat bucket once PER GROUP; silently double-
+    # stepping with two lrs would corrupt training — refuse at attach time
+    import pytest
+    import torch
+
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.parallel.reducer import Reducer
+
+    m1, m2 = toy_model(8, 1), toy_model(8, 1)
+    opt = FusedSGD([{"params": m1.parameters(), "lr": 1e-3},
+                    {"params": m2.parameters(), "lr": 1e-4}], lr=1e-3)
+    red = Reducer(list(m1.parameters()) + list(m2.parameters()), comm=None)
+    with pytest.raises(ValueError, match="single param_group"):
+        opt.attach_reducer(red)
