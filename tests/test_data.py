@@ -126,3 +126,20 @@ def test_perm_index_is_bijection_fuzz():
         assert len(seen) == n and min(seen) == 0 and max(seen) == n - 1
 
     check()
+
+
+def test_sharded_sampler_drop_last_matches_torch():
+    import torch
+    import torch.utils.data as tud
+
+    from mi355x_ddp.data import ShardedSampler, ToyDataset
+
+    for n, world in ((37, 4), (40, 4), (17, 3), (8, 8)):
+        ds = ToyDataset(n, seed=1)
+        for rank in range(world):
+            ours = ShardedSampler(ds, num_replicas=world, rank=rank,
+                                  shuffle=False, drop_last=True)
+            ref = tud.DistributedSampler(ds, num_replicas=world, rank=rank,
+                                         shuffle=False, drop_last=True)
+            assert list(ours) == list(ref), (n, world, rank)
+            assert len(ours) == len(ref)
