@@ -173,3 +173,17 @@ def test_unknown_engine_rejected():
     dl = prepare_dataloader(ds, 32)
     with pytest.raises(ValueError, match="unknown engine"):
         Trainer(model, dl, opt, "cpu", 1, wrap_ddp=False, engine="presistent")
+
+
+def test_framework_snapshot_loads_into_plain_nn_linear(tmp_chdir):
+    """Reverse interop: a snapshot WE wrote restores into the reference's
+    model type with the reference's loading pattern (torch.load +
+    load_state_dict on nn.Linear(20,1); ref multigpu_torchrun.py:36-41) —
+    the forward direction is test_reference_written_snapshot_loads."""
+    tr = _make(snapshot_path="snapshot.pt")
+    tr._save_snapshot(2)
+    snap = torch.load("snapshot.pt", weights_only=True)
+    ref = torch.nn.Linear(20, 1)
+    ref.load_state_dict(snap["MODEL_STATE"])  # exact key/shape match
+    assert snap["EPOCHS_RUN"] == 2
+    assert torch.equal(ref.weight.data, tr._unwrapped().weight.data)
