@@ -159,9 +159,13 @@ class Reducer:
 
     # -- backward-side machinery ----------------------------------------
     def _make_hook(self, p: torch.nn.Parameter):
+        # bucket/index resolved at registration time (_param_index is
+        # final by then) — the hook body runs once per param per step
+        b, idx = self._param_index[p]
+        grad_views = self.grad_views  # fixed for the Reducer's lifetime
+
         def hook(param: torch.nn.Parameter) -> None:
-            b, idx = self._param_index[param]
-            if self.grad_views:
+            if grad_views:
                 g = param.grad
                 view = b.grad_view(idx)
                 if g is not None and g.data_ptr() != view.data_ptr():
