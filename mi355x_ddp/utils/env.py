@@ -16,6 +16,16 @@ MI355X_PREFETCH         0 = disable the Trainer's copy-stream H2D
 MI355X_P2P              0 = disable the xGMI mesh all-reduce in bench.py
                         (default on; mesh is only adopted after all ranks
                         agree it set up and cross-validated)
+MI355X_EPOCH_BLOCK      epochs gathered per epoch_shard_multi launch in
+                        bench.py, letting the multistep launch span epoch
+                        boundaries (default: fills the engine's max_defer
+                        window; 1 = per-epoch gathers)
+MI355X_ENGINE           fast-engine selection for the entrypoint scripts
+                        (hooks default | auto | fused | persistent |
+                        graph; silently falls back to hooks when the
+                        stage does not qualify)
+MI355X_DTYPE            bf16 = cast the model in multigpu.py (BASELINE
+                        config 2); unset = the reference's fp32
 
 RCCL's own tuning envs (NCCL_ALGO, NCCL_PROTO, NCCL_MIN/MAX_NCHANNELS)
 pass straight through to the large-bucket collective path — the knobs
