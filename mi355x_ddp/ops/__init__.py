@@ -127,12 +127,21 @@ class _HipCEFn(torch.autograd.Function):
 
 
 def cross_entropy(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
-    """torch.nn.CrossEntropyLoss()(output, target) with probability targets.
+    """torch.nn.CrossEntropyLoss()(output, target): probability targets
+    (the reference's usage — float targets of the output's shape) AND the
+    common class-INDEX form (integer targets of shape [B]), which torch
+    dispatches on dtype; index targets run the same kernels through their
+    exact one-hot probability equivalent.
 
     Note the reference's toy case is degenerate (C=1 -> loss == 0, grads
     == 0; SURVEY §2.1 'Degenerate loss') — both paths reproduce torch's
     exact semantics for it.
     """
+    if not target.is_floating_point():
+        if output.is_cuda:
+            t = F.one_hot(target.long(), output.shape[-1]).to(output.dtype)
+            return _HipCEFn.apply(output, t)
+        return F.cross_entropy(output, target)
     if output.is_cuda:
         return _HipCEFn.apply(output, target.to(output.dtype))
     return F.cross_entropy(output, target.to(output.dtype))

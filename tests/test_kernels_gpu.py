@@ -253,3 +253,20 @@ def test_epoch_shard_different_epochs_differ():
     a, _ = ops.ext().epoch_shard(X, T, 1, 0, 1)
     b, _ = ops.ext().epoch_shard(X, T, 2, 0, 1)
     assert not torch.equal(a, b)
+
+
+def test_ce_class_index_targets_gpu():
+    """Index-target CE on device: loss + input grads match torch's
+    index-target CE (the kernels see the exact one-hot equivalent)."""
+    from mi355x_ddp import ops
+
+    torch.manual_seed(9)
+    y = torch.randn(16, 50, device="cuda", requires_grad=True)
+    idx = torch.randint(0, 50, (16,), device="cuda")
+    loss = ops.cross_entropy(y, idx)
+    want = torch.nn.functional.cross_entropy(y.detach(), idx)
+    assert torch.allclose(loss, want, atol=1e-5)
+    loss.backward()
+    y2 = y.detach().clone().requires_grad_(True)
+    torch.nn.functional.cross_entropy(y2, idx).backward()
+    assert torch.allclose(y.grad, y2.grad, atol=1e-5)

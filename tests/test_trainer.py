@@ -187,3 +187,20 @@ def test_framework_snapshot_loads_into_plain_nn_linear(tmp_chdir):
     ref.load_state_dict(snap["MODEL_STATE"])  # exact key/shape match
     assert snap["EPOCHS_RUN"] == 2
     assert torch.equal(ref.weight.data, tr._unwrapped().weight.data)
+
+
+def test_ce_loss_class_index_targets_match_torch():
+    """The common CrossEntropyLoss form (integer class indices, shape [B])
+    works as a drop-in: same loss and same input gradients as torch."""
+    from mi355x_ddp import ops
+
+    torch.manual_seed(9)
+    y = torch.randn(8, 5, requires_grad=True)
+    idx = torch.randint(0, 5, (8,))
+    loss = ops.cross_entropy(y, idx)
+    want = torch.nn.functional.cross_entropy(y.detach(), idx)
+    assert torch.allclose(loss, want, atol=1e-6)
+    loss.backward()
+    y2 = y.detach().clone().requires_grad_(True)
+    torch.nn.functional.cross_entropy(y2, idx).backward()
+    assert torch.allclose(y.grad, y2.grad, atol=1e-6)
