@@ -496,6 +496,9 @@ __global__ void k_ce_fwd(const T* __restrict__ Y, const T* __restrict__ Tg,
 std::vector<torch::Tensor> ce_fwd(torch::Tensor y, torch::Tensor t) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
+  TORCH_CHECK(tc.sizes() == yc.sizes(),
+              "ce: probability targets must match the output shape (got ",
+              tc.sizes(), " vs ", yc.sizes(), ")");
   const int B = (int)yc.size(0), C = (int)yc.size(1);
   auto f32opt = yc.options().dtype(at::kFloat);
   auto probs = at::empty({B, C}, f32opt);
@@ -568,6 +571,8 @@ __global__ void k_mse_fwd(const T* __restrict__ Y, const T* __restrict__ Tg,
 torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
+  TORCH_CHECK(tc.numel() == yc.numel(), "mse: target numel ", tc.numel(),
+              " != output numel ", yc.numel());
   const int64_t n = yc.numel();
   auto loss = at::zeros({}, yc.options().dtype(at::kFloat));
   const int blocks = (int)std::min<int64_t>(cdiv(n, 256), 2048);
