@@ -99,6 +99,13 @@ class _HipLinearFn(torch.autograd.Function):
 
 def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None):
     if x.is_cuda:
+        if x.dim() != 2:
+            # nn.Linear semantics for arbitrary leading dims: flatten to
+            # the kernels' 2-D contract, restore after (reshape is
+            # autograd-transparent)
+            lead = x.shape[:-1]
+            y = _HipLinearFn.apply(x.reshape(-1, x.shape[-1]), w, b)
+            return y.reshape(*lead, w.shape[0])
         return _HipLinearFn.apply(x, w, b)
     return F.linear(x, w, b)
 

@@ -270,3 +270,23 @@ def test_ce_class_index_targets_gpu():
     y2 = y.detach().clone().requires_grad_(True)
     torch.nn.functional.cross_entropy(y2, idx).backward()
     assert torch.allclose(y.grad, y2.grad, atol=1e-5)
+
+
+def test_linear_nd_leading_dims():
+    """HipLinear accepts [B, T, K] like nn.Linear: forward and grads match
+    torch (flatten-to-2D around the MFMA kernels)."""
+    from mi355x_ddp import ops
+
+    torch.manual_seed(12)
+    x = torch.randn(4, 7, 20, device="cuda", requires_grad=True)
+    w = torch.randn(3, 20, device="cuda", requires_grad=True)
+    b = torch.randn(3, device="cuda", requires_grad=True)
+    y = ops.linear(x, w, b)
+    assert y.shape == (4, 7, 3)
+    y.sum().backward()
+    x2 = x.detach().clone().requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+    torch.nn.functional.linear(x2, w2, b2).sum().backward()
+    for a, r in ((x.grad, x2.grad), (w.grad, w2.grad), (b.grad, b2.grad)):
+        assert torch.allclose(a, r, atol=1e-4, rtol=1e-4)

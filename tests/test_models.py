@@ -31,3 +31,19 @@ def test_vit_tiny_trains_cpu():
         losses.append(float(loss.detach()))
     assert all(torch.isfinite(torch.tensor(losses)))
     assert losses[-1] < losses[0]
+
+
+def test_hip_linear_nd_input_cpu():
+    # nn.Linear drop-in for arbitrary leading dims (transformer-style
+    # [B, T, K] inputs) — CPU path
+    import torch
+
+    from mi355x_ddp.models.toy import HipLinear
+
+    torch.manual_seed(4)
+    m = HipLinear(20, 3)
+    x = torch.randn(2, 5, 20)
+    y = m(x)
+    assert y.shape == (2, 5, 3)
+    ref = torch.nn.functional.linear(x, m.weight, m.bias)
+    assert torch.allclose(y, ref, atol=1e-6)
