@@ -37,7 +37,9 @@ class ToyFusedStep:
     def __init__(self, model: HipLinear, comm=None, lr: float = 1e-3,
                  use_mse: bool = True, reducer: Optional[Reducer] = None,
                  track_loss: bool = False):
-        assert isinstance(model, HipLinear) and model.out_features == 1
+        assert (isinstance(model, HipLinear) and model.out_features == 1
+                and model.bias is not None), \
+            "toy engines need HipLinear(K, 1) with a bias"
         self.model = model
         self.comm = comm
         self.lr = lr

@@ -146,6 +146,7 @@ class Trainer:
         from .parallel import FusedSGD
         m = self.model
         if not (isinstance(m, HipLinear) and m.out_features == 1
+                and m.bias is not None
                 and self.device.type == "cuda"
                 and self.loss_fn is ops.mse_loss
                 and isinstance(self.optimizer, FusedSGD)
