@@ -269,6 +269,10 @@ def main():
     data = DeviceData(args.dataset, rank, world, args.batch, device,
                       dtype=dtype if use_cuda else torch.float32)
     spe = data.steps_per_epoch
+    if spe < 1:
+        raise SystemExit(
+            f"--dataset {args.dataset} yields {data.per_rank} samples/rank "
+            f"at world {world}: smaller than one --batch {args.batch}")
 
     step_fn = flush_fn = engine_obj = None
     eng_bind = eng_step_shard = None
