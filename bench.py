@@ -355,11 +355,14 @@ def main():
     # -- timed region: exactly K steps (any deferred launches are flushed
     #    INSIDE the bracket — all K steps' work executes before the
     #    closing barrier+synchronize) ------------------------------------
+    launches0 = getattr(engine_obj, "launch_count", None)
     t0 = time.perf_counter()
     run_steps(args.warmup, args.steps)
     flush_fn()
     barrier()
     elapsed = time.perf_counter() - t0
+    launches_timed = (getattr(engine_obj, "launch_count", 0) - launches0
+                      if launches0 is not None else None)
     comm_check()
 
     # max over ranks (gloo all-reduce of the scalar)
@@ -419,8 +422,22 @@ def main():
                 "loss": args.loss,
                 "epoch_block": eb,
                 "p50_step_ms": p50_ms,
+                "launches_timed": launches_timed,
             },
         }
+        # Transparency for short runs (the driver defaults to --steps 20):
+        # the deferred engine amortizes ~2.7 us of launch overhead over up
+        # to max_defer steps, so a timed region spanning only a couple of
+        # launches underestimates the steady state (measured at 30k steps,
+        # profiles/README.md r01q). Say so in the JSON instead of leaving
+        # an apparent builder-vs-driver discrepancy.
+        if args.steps < 2000:
+            result["config"]["steps_note"] = (
+                f"short run: {args.steps} timed steps across "
+                f"{launches_timed if launches_timed is not None else '?'} "
+                "kernel launch(es) — launch overhead not amortized; "
+                "steady-state throughput (>=30000 steps) is higher "
+                "(see profiles/README.md)")
         print(json.dumps(result))
     if world > 1:
         dist.destroy_process_group()

@@ -112,9 +112,12 @@ class PersistentToyStep(ToyFusedStep):
         lr, max_defer = self.lr, self.max_defer
         mesh = self._mesh
 
+        self.launch_count = 0  # kernel launches issued (bench transparency)
+
         def launch(xall, tall, B):
             """One deferred-run launch: world-1 multistep, or the mesh
             variant with an in-kernel all-reduce per step."""
+            self.launch_count += 1
             if mesh is None:
                 ext_mod.toy_multistep(xall, tall, flat_param, loss_out,
                                       use_mse, w_off, b_off, lr, B)
@@ -235,7 +238,11 @@ class GraphedToyStep(ToyFusedStep):
             with torch.cuda.graph(self._graph):
                 super().step(self._x_static, self._t_static)
         except Exception as e:  # capture unsupported -> eager fallback
-            print(f"[mi355x_ddp] hipGraph capture failed ({e}); running eager")
+            import warnings
+            warnings.warn(
+                f"[mi355x_ddp] hipGraph capture failed ({e!r}); "
+                "GraphedToyStep running eager (correct, slower)",
+                RuntimeWarning, stacklevel=2)
             self._graph = False
             torch.cuda.synchronize()
 

@@ -145,6 +145,16 @@ def cross_entropy(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
     exact semantics for it.
     """
     if not target.is_floating_point():
+        if (target < 0).any():
+            # torch maps negative indices to ignore_index semantics
+            # (default -100: skip the sample, renormalize the mean). The
+            # GPU kernels don't implement that, and letting the CPU
+            # fallback silently honor it would make the two paths diverge
+            # on identical inputs — reject on BOTH instead.
+            raise ValueError(
+                "mi355x_ddp.ops.cross_entropy: negative class indices "
+                "(torch's ignore_index) are not supported; filter ignored "
+                "samples out before the loss")
         if output.is_cuda:
             t = F.one_hot(target.long(), output.shape[-1]).to(output.dtype)
             return _HipCEFn.apply(output, t)

@@ -571,8 +571,13 @@ __global__ void k_mse_fwd(const T* __restrict__ Y, const T* __restrict__ Tg,
 torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
-  TORCH_CHECK(tc.numel() == yc.numel(), "mse: target numel ", tc.numel(),
-              " != output numel ", yc.numel());
+  // Full-size equality, like ce_fwd: equal-NUMEL shape mismatches (e.g.
+  // [B,1] vs [B]) would compute elementwise here while torch's
+  // F.mse_loss broadcasts the pair to [B,B] — a silent numeric
+  // divergence. The reference's usage always has matching shapes, so a
+  // hard error is safe.
+  TORCH_CHECK(tc.sizes() == yc.sizes(), "mse: target shape ", tc.sizes(),
+              " != output shape ", yc.sizes());
   const int64_t n = yc.numel();
   auto loss = at::zeros({}, yc.options().dtype(at::kFloat));
   const int blocks = (int)std::min<int64_t>(cdiv(n, 256), 2048);

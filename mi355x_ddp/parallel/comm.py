@@ -173,12 +173,11 @@ class P2pMeshComm:
         nccl-only — still a transport independent of this mesh)."""
         probe = torch.arange(24, dtype=torch.float32, device=self._device)
         probe = probe * (self.rank + 1)
-        ref = probe.cpu()
-        try:
-            dist.all_reduce(ref, op=dist.ReduceOp.SUM)
-        except RuntimeError:
+        if _cpu_collectives_ok():
+            ref = probe.cpu()
+        else:
             ref = probe.clone()  # device tensor for an nccl-only group
-            dist.all_reduce(ref, op=dist.ReduceOp.SUM)
+        dist.all_reduce(ref, op=dist.ReduceOp.SUM)
         ref = ref.cpu() / self.world
         self._mesh.all_reduce_avg_inline(probe)
         torch.cuda.synchronize()
@@ -191,8 +190,29 @@ class P2pMeshComm:
     def check(self) -> None:
         self._mesh.check()
 
+    # payloads at or under this ride the device-side mesh; larger ones the
+    # base transport. The threshold is a constant, so routing is identical
+    # across ranks WHENEVER the call-site tensor sizes agree — which the
+    # reducer guarantees (identical bucket layout on every rank) and
+    # MI355X_DEBUG_SYNC=1 verifies per call (all_ranks_same_size).
+    MESH_MAX_NUMEL = 64
+
+    def _assert_uniform_size(self, t: torch.Tensor) -> None:
+        lohi = torch.tensor([t.numel(), -t.numel()], dtype=torch.int64)
+        if not _cpu_collectives_ok():
+            lohi = lohi.to(_this_rank_device())
+        dist.all_reduce(lohi, op=dist.ReduceOp.MIN)
+        lo, hi = int(lohi[0]), -int(lohi[1])
+        if lo != hi:
+            raise RuntimeError(
+                f"P2pMeshComm size-routing divergence: this rank reduces "
+                f"{t.numel()} elements but the world spans [{lo}, {hi}] — "
+                "ranks would take different transports and hang")
+
     def all_reduce_avg_inline(self, t: torch.Tensor) -> None:
-        if t.numel() <= 64:
+        if os.environ.get("MI355X_DEBUG_SYNC") == "1":
+            self._assert_uniform_size(t)
+        if t.numel() <= self.MESH_MAX_NUMEL:
             self._mesh.all_reduce_avg_inline(t)
         else:
             self.base.all_reduce_avg_inline(t)
@@ -211,6 +231,36 @@ class P2pMeshComm:
         self.base.barrier()
 
 
+def _cpu_collectives_ok() -> bool:
+    """Does the default process group accept CPU tensors? Probed from the
+    group's per-device backend map (e.g. "cpu:gloo,cuda:nccl") instead of
+    retrying on RuntimeError — a transient gloo failure must surface, not
+    be misread as 'backend lacks CPU support'. Deterministic across ranks:
+    every rank sees the same backend config, so every rank takes the same
+    transport branch."""
+    try:
+        config = str(dist.get_backend_config())
+    except Exception:
+        config = str(dist.get_backend())
+    for entry in config.split(","):
+        dev, _, backend = entry.partition(":")
+        if not backend:  # bare backend name, applies to all devices
+            return dev in ("gloo", "mpi")
+        if dev == "cpu":
+            return backend in ("gloo", "mpi")
+    return False  # no cpu entry: cuda-only (nccl) group
+
+
+def _this_rank_device() -> torch.device:
+    """The CUDA device belonging to THIS local rank. Never bare .cuda():
+    for a user-initialized group our ddp_setup (which calls
+    torch.cuda.set_device) may not have run, and cuda:0 from every local
+    rank would collide in NCCL."""
+    if "LOCAL_RANK" in os.environ:
+        return torch.device("cuda", int(os.environ["LOCAL_RANK"]))
+    return torch.device("cuda", torch.cuda.current_device())
+
+
 def all_ranks_agree(ok: bool) -> bool:
     """World-wide MIN over a per-rank success flag (gloo/default group).
 
@@ -220,14 +270,11 @@ def all_ranks_agree(ok: bool) -> bool:
     transport decision must therefore be agreed by all ranks — call this
     at the same program point on every rank."""
     t = torch.tensor([1 if ok else 0])
-    try:
-        dist.all_reduce(t, op=dist.ReduceOp.MIN)
-    except RuntimeError:
+    if not _cpu_collectives_ok():
         # nccl-only default group (user-initialized): agree via a device
-        # tensor instead. Deterministic across ranks — every rank sees the
-        # same group backend, so every rank takes the same branch.
-        t = t.cuda()
-        dist.all_reduce(t, op=dist.ReduceOp.MIN)
+        # tensor on THIS rank's device instead.
+        t = t.to(_this_rank_device())
+    dist.all_reduce(t, op=dist.ReduceOp.MIN)
     return bool(t.item())
 
 
@@ -275,11 +322,20 @@ def build_gpu_comm(device: torch.device, want_mesh: bool = True,
 
 def create_comm(device: torch.device):
     """Pick the communicator for this process, or None when world size is 1
-    (single-process runs use the same reducer with no collectives)."""
+    (single-process runs use the same reducer with no collectives).
+
+    GPU worlds go through the SAME hang-safe ladder as bench/the fast
+    engines (build_gpu_comm): a rank-asymmetric RcclComm init failure
+    downgrades ALL ranks to gloo together instead of leaving the world
+    with mismatched transports (the exact hang all_ranks_agree documents).
+    The mesh layer is skipped — the generic reducer only issues bucket
+    all_reduce_avg calls, which the mesh routes to the base transport
+    anyway."""
     if not (dist.is_available() and dist.is_initialized()):
         return None
     if dist.get_world_size() == 1:
         return None
     if device.type == "cuda":
-        return RcclCommAdapter(device)
+        comm, _kind = build_gpu_comm(device, want_mesh=False)
+        return comm
     return GlooComm()

@@ -45,12 +45,18 @@ class DDP(torch.nn.Module):
                                  "DDP's single-process-single-device mode)")
             want = torch.device("cuda", device_ids[0]) \
                 if isinstance(device_ids[0], int) else torch.device(device_ids[0])
-            for p in module.parameters():
-                if p.device != want:
+            if want.type == "cuda" and want.index is None:
+                # indexless "cuda" (stock DDP accepts it): means the current
+                # device, like tensor.cuda() would
+                want = torch.device("cuda", torch.cuda.current_device())
+            for name, p in module.named_parameters():
+                have = p.device
+                if have.type == "cuda" and have.index is None:
+                    have = torch.device("cuda", torch.cuda.current_device())
+                if (have.type, have.index) != (want.type, want.index):
                     raise ValueError(
-                        f"model is on {p.device} but device_ids names {want}; "
-                        "move the model first (model.to(device))")
-                break
+                        f"parameter {name} is on {p.device} but device_ids "
+                        f"names {want}; move the model first (model.to(device))")
         self.module = module
         params = list(module.parameters())
         device = params[0].device if params else torch.device("cpu")
