@@ -4,6 +4,15 @@
 #include "ops.h"
 #include "p2p_mesh.h"
 #include "rccl_comm.h"
+#include "reducer_core.h"
+
+namespace mi355x {
+// autograd_ops.hip
+torch::Tensor linear_autograd(torch::Tensor x, torch::Tensor w,
+                              c10::optional<torch::Tensor> b);
+torch::Tensor mse_autograd(torch::Tensor y, torch::Tensor t);
+torch::Tensor ce_autograd(torch::Tensor y, torch::Tensor t);
+}  // namespace mi355x
 
 namespace py = pybind11;
 
@@ -65,6 +74,33 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("check", &mi355x::P2pMesh::check)
       .def_property_readonly("rank", &mi355x::P2pMesh::rank)
       .def_property_readonly("world", &mi355x::P2pMesh::world);
+
+  // C++ autograd Functions (whole fwd/bwd chain stays out of Python)
+  m.def("linear_autograd", &mi355x::linear_autograd, py::arg("x"),
+        py::arg("w"), py::arg("bias") = c10::nullopt);
+  m.def("mse_autograd", &mi355x::mse_autograd);
+  m.def("ce_autograd", &mi355x::ce_autograd);
+
+  py::class_<mi355x::ReducerCore, std::shared_ptr<mi355x::ReducerCore>>(
+      m, "ReducerCore")
+      .def(py::init([](std::vector<std::vector<torch::Tensor>> params,
+                       std::vector<std::vector<torch::Tensor>> views,
+                       std::vector<torch::Tensor> flats,
+                       mi355x::RcclComm* comm) {
+             return std::make_shared<mi355x::ReducerCore>(
+                 std::move(params), std::move(views), std::move(flats), comm);
+           }),
+           py::arg("bucket_params"), py::arg("bucket_views"),
+           py::arg("bucket_flat_grads"), py::arg("comm") = nullptr,
+           // the core holds a non-owning RcclComm*: keep the comm alive as
+           // long as the core (nurse the adapter's lifetime from C++ side)
+           py::keep_alive<1, 5>())
+      .def("attach_hooks", &mi355x::ReducerCore::attach_hooks)
+      .def("detach_hooks", &mi355x::ReducerCore::detach_hooks)
+      .def("finalize", &mi355x::ReducerCore::finalize,
+           py::call_guard<py::gil_scoped_release>())
+      .def("set_skip_comm", &mi355x::ReducerCore::set_skip_comm)
+      .def_property_readonly("steps", &mi355x::ReducerCore::steps);
 
   py::class_<mi355x::RcclComm>(m, "RcclComm")
       .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),

@@ -143,12 +143,52 @@ class Reducer:
                 self._param_index[p] = (b, i)
                 p._mi355x_bucket = b.index  # used by FusedSGD
 
-        self._hooks = [
-            p.register_post_accumulate_grad_hook(self._make_hook(p))
-            for b in self.buckets for p in b.params
-        ]
+        # Hook transport: the C++ ReducerCore installs post-hooks directly
+        # on the AccumulateGrad nodes — ready-counting and the bucket
+        # all-reduce launch run on the autograd engine thread without the
+        # GIL (torch's own reducer.cpp is C++ for the same
+        # reason). Engaged on the GPU views path whenever the communicator
+        # is native RCCL (or absent); the Python hook path remains for
+        # CPU/gloo testing and copy-mode. MI355X_CPP_HOOKS=0 forces the
+        # Python path (used by the GPU parity tests).
+        self._core = None
+        self._hooks = []
+        self._skip_comm = False  # set by DDP.no_sync() during accumulation
+        use_core = (self.grad_views and device.type == "cuda"
+                    and os.environ.get("MI355X_CPP_HOOKS", "1") != "0"
+                    and ops.has_ext())
+        core_comm = None
+        if use_core and comm is not None:
+            from .comm import P2pMeshComm, RcclCommAdapter
+            base = comm.base if isinstance(comm, P2pMeshComm) else comm
+            if isinstance(base, RcclCommAdapter):
+                core_comm = base._comm
+            else:
+                use_core = False  # gloo shadow: Python hooks mirror it
+        if use_core:
+            self._core = ops.ext().ReducerCore(
+                [list(b.params) for b in self.buckets],
+                [[b.grad_view(i) for i in range(len(b.params))]
+                 for b in self.buckets],
+                [b.flat_grad for b in self.buckets],
+                core_comm)
+            self._core.attach_hooks()
+        else:
+            self._hooks = [
+                p.register_post_accumulate_grad_hook(self._make_hook(p))
+                for b in self.buckets for p in b.params
+            ]
         self._next_launch = 0
-        self.skip_comm = False  # set by DDP.no_sync() during accumulation
+
+    @property
+    def skip_comm(self) -> bool:
+        return self._skip_comm
+
+    @skip_comm.setter
+    def skip_comm(self, v: bool) -> None:
+        self._skip_comm = v
+        if self._core is not None:
+            self._core.set_skip_comm(v)
 
     # -- wrap-time state sync --------------------------------
     def broadcast_params(self, root: int = 0) -> None:
@@ -234,6 +274,9 @@ class Reducer:
     def finalize(self) -> None:
         """Called after loss.backward(): launch stragglers, fence compute on
         the comm stream, and reset per-step state."""
+        if self._core is not None:
+            self._core.finalize()  # C++: stragglers + fence + reset, no GIL
+            return
         for b in self.buckets[self._next_launch:]:
             if not self.grad_views and not self.skip_comm:
                 self._flatten_bucket(b)
@@ -252,5 +295,7 @@ class Reducer:
         return [(b.flat_param, b.flat_grad) for b in self.buckets]
 
     def detach_hooks(self) -> None:
+        if self._core is not None:
+            self._core.detach_hooks()
         for h in self._hooks:
             h.remove()

@@ -19,6 +19,8 @@ ext = CUDAExtension(
         "mi355x_ddp/ops/csrc/kernels.hip",
         "mi355x_ddp/ops/csrc/rccl_comm.hip",
         "mi355x_ddp/ops/csrc/p2p_mesh.hip",
+        "mi355x_ddp/ops/csrc/reducer_core.hip",
+        "mi355x_ddp/ops/csrc/autograd_ops.hip",
         "mi355x_ddp/ops/csrc/bindings.hip",
     ],
     libraries=["rccl"],

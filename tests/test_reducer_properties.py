@@ -123,3 +123,17 @@ def test_fused_sgd_rejects_multiple_groups_on_attach():
     red = Reducer(list(m1.parameters()) + list(m2.parameters()), comm=None)
     with pytest.raises(ValueError, match="single param_group"):
         opt.attach_reducer(red)
+
+
+def test_cpp_core_not_engaged_on_cpu():
+    # the C++ ReducerCore is the GPU hook path; CPU (gloo test tier) must
+    # keep the Python hooks regardless of the env default
+    from mi355x_ddp.parallel.reducer import Reducer
+
+    m = nn.Linear(8, 4)
+    red = Reducer(list(m.parameters()), comm=None, grad_views=True)
+    assert red._core is None and red._hooks
+    # skip_comm stays a plain flag on the Python path
+    red.skip_comm = True
+    assert red.skip_comm is True
+    red.skip_comm = False
