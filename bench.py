@@ -36,10 +36,14 @@ def parse_args():
     p.add_argument("--steps", type=int, default=2000)
     p.add_argument("--warmup", type=int, default=200)
     p.add_argument("--engine",
-                   choices=["persistent", "fused", "graph", "autograd"],
+                   choices=["persistent", "fused", "graph", "autograd",
+                            "autograd-graph"],
                    default="persistent",
                    help="persistent = multi-step kernel at world 1 "
-                        "(falls back to fused when a comm exists)")
+                        "(falls back to fused when a comm exists); "
+                        "autograd = the generic hook/reducer path; "
+                        "autograd-graph = the generic path captured in "
+                        "one hipGraph and replayed (whole-step capture)")
     p.add_argument("--batch", type=int, default=32, help="batch per rank")
     p.add_argument("--lr", type=float, default=1e-3)
     p.add_argument("--dataset", type=int, default=2048)
@@ -202,12 +206,18 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32, use_mse=True):
     model = toy_model(20, 1).to(device=device, dtype=dtype)
 
     noflush = lambda: None  # noqa: E731
-    if kind == "autograd":
+    if kind in ("autograd", "autograd-graph"):
         engine = DDP(model, comm=comm)
         opt = FusedSGD(model.parameters(), lr=lr)
         opt.attach_reducer(engine.reducer)
 
         loss_fn = ops.mse_loss if use_mse else ops.cross_entropy
+
+        if kind == "autograd-graph" and device.type == "cuda":
+            from mi355x_ddp.engine import GraphedAutogradStep
+            gs = GraphedAutogradStep(engine, loss_fn, opt,
+                                     finalize=engine.finalize_backward)
+            return gs.step, noflush, None
 
         def step(x, t):
             loss = loss_fn(engine(x), t)
@@ -414,7 +424,7 @@ def main():
                 "parallelism": f"dp{world}",
                 "comm": comm_kind,
                 "engine": ("autograd-cpu" if not use_cuda else
-                           "autograd" if engine_obj is None else
+                           args.engine if engine_obj is None else
                            {"PersistentToyStep": "persistent",
                             "GraphedToyStep": "graph",
                             "ToyFusedStep": "fused"}[

@@ -211,6 +211,96 @@ class PersistentToyStep(ToyFusedStep):
         self.step_shard = step_shard
 
 
+class GraphedAutogradStep:
+    """The GENERIC training step — forward, loss, autograd backward with
+    the ReducerCore hooks, bucket all-reduce, fused SGD — captured in ONE
+    hipGraph and replayed (torch "whole-network capture").
+
+    This is what makes the arbitrary-model path fast on MI355X: eager, the
+    step pays ~10 kernel launches plus autograd-engine and Python dispatch
+    overhead (~100 us for the toy model, profiles/README.md r02); replayed,
+    it pays two tiny input copies plus the graph-replay floor (~10-16 us,
+    MI355X_MICROARCH price list). The step is capture-safe BY CONSTRUCTION
+    here: gradients are views into the reducer's static flat buckets,
+    FusedSGD updates the static flat pairs, and the C++ hook trampoline
+    launches collectives with stream-event edges capture can record.
+
+    Requirements: static shapes (same batch size every step) and a model
+    without data-dependent control flow — the reference workloads (toy
+    Linear, ResNet-50) qualify. Capture failure falls back to eager with a
+    warning; results are unchanged either way.
+    """
+
+    def __init__(self, model, loss_fn, optimizer, finalize=None,
+                 warmup_steps: int = 3, zero_grad: bool = False):
+        self.model = model            # DDP-wrapped or bare
+        self.loss_fn = loss_fn
+        self.optimizer = optimizer
+        self._finalize = finalize or (lambda: None)
+        self.warmup_steps = warmup_steps
+        # zero_grad=True: the optimizer does not fold grad zeroing into its
+        # step (i.e. it is not a bucket-attached FusedSGD) — zero INSIDE
+        # the captured step, set_to_none=False so the grad buffers stay
+        # static (a capture requirement, and it keeps the bucket views).
+        self._zero = zero_grad
+        self._graph = None
+        self._x = self._t = None
+        self.loss = None              # captured loss tensor (replay target)
+
+    def _eager_step(self, x, t):
+        if self._zero:
+            self.optimizer.zero_grad(set_to_none=False)
+        loss = self.loss_fn(self.model(x), t)
+        loss.backward()
+        self._finalize()
+        self.optimizer.step()
+        return loss
+
+    def _capture(self, x, t):
+        self._x = x.clone()
+        self._t = t.clone()
+        # torch's capture recipe: warm up on a side stream first (allocator
+        # + autograd engine touch lazily-initialized state off-capture)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(self.warmup_steps):
+                self._eager_step(self._x, self._t)
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        try:
+            g = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(g):
+                self.loss = self._eager_step(self._x, self._t)
+            self._graph = g
+        except Exception as e:
+            import warnings
+            warnings.warn(
+                f"[mi355x_ddp] whole-step hipGraph capture failed ({e!r}); "
+                "GraphedAutogradStep running eager (correct, slower)",
+                RuntimeWarning, stacklevel=2)
+            self._graph = False
+            torch.cuda.synchronize()
+
+    def step(self, x, t):
+        if self._graph is None:
+            self._capture(x, t)
+            if self._graph:
+                self._graph.replay()  # capture records, replay executes:
+                # this call still performs exactly one (post-warmup) step
+            else:
+                self._eager_step(x, t)
+            return
+        if self._graph is False or x.shape != self._x.shape:
+            # eager fallback; shape mismatch = a ragged final batch — run
+            # it eager rather than replaying a graph of the wrong shape
+            self._eager_step(x, t)
+            return
+        self._x.copy_(x, non_blocking=True)
+        self._t.copy_(t, non_blocking=True)
+        self._graph.replay()
+
+
 class GraphedToyStep(ToyFusedStep):
     """ToyFusedStep captured in a hipGraph.
 

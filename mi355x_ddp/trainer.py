@@ -81,14 +81,18 @@ class Trainer:
         engine: str = "hooks",
     ) -> None:
         """engine: "hooks" (default — the generic autograd+reducer path,
-        reference semantics for arbitrary models), or "fused"/"persistent"
-        /"graph"/"auto" to run the toy fast path (single-launch fused step /
-        multi-step deferred kernel / hipGraph replay) when the
-        model+loss+optimizer qualify; silently falls back to hooks
-        otherwise. Unknown values raise."""
-        if engine not in ("hooks", "auto", "fused", "persistent", "graph"):
-            raise ValueError(f"unknown engine {engine!r} (hooks | auto | "
-                             "fused | persistent | graph)")
+        reference semantics for arbitrary models), "hooks-graph" (the same
+        generic path captured in one hipGraph and replayed — fast for any
+        static-shape model; falls back to eager hooks if capture fails or
+        off-GPU), or "fused"/"persistent"/"graph"/"auto" to run the toy
+        fast path (single-launch fused step / multi-step deferred kernel /
+        hipGraph replay) when the model+loss+optimizer qualify; silently
+        falls back to hooks otherwise. Unknown values raise."""
+        if engine not in ("hooks", "hooks-graph", "auto", "fused",
+                          "persistent", "graph"):
+            raise ValueError(f"unknown engine {engine!r} (hooks | "
+                             "hooks-graph | auto | fused | persistent | "
+                             "graph)")
         if gpu_id is None:
             gpu_id = int(os.environ.get("LOCAL_RANK", 0))
         self.gpu_id = gpu_id
@@ -126,7 +130,8 @@ class Trainer:
 
         self._distributed = False
         self._engine = None
-        if engine != "hooks" and self._try_fast_engine(engine):
+        if engine not in ("hooks", "hooks-graph") \
+                and self._try_fast_engine(engine):
             pass  # _run_batch drives self._engine
         elif wrap_ddp and dist.is_available() and dist.is_initialized():
             # world 1 wraps too: the engine still provides flat buckets and
@@ -136,6 +141,22 @@ class Trainer:
             self._distributed = True
             if isinstance(self.optimizer, FusedSGD):
                 self.optimizer.attach_reducer(self.model.reducer)
+            # A bucket-attached FusedSGD zeroes the flat grads inside its
+            # step kernel; any other optimizer needs an explicit zero or
+            # gradients would accumulate across steps.
+            self._needs_zero = not (
+                isinstance(self.optimizer, FusedSGD)
+                and self.optimizer._flat_pairs is not None)
+            if engine == "hooks-graph" and self.device.type == "cuda":
+                # whole-step hipGraph capture of the generic path: one
+                # replay per step instead of ~10 launches + autograd
+                # overhead. Eager-fallback (with a warning) on capture
+                # failure — training results are identical either way.
+                from .engine import GraphedAutogradStep
+                self._engine = GraphedAutogradStep(
+                    self.model, self.loss_fn, self.optimizer,
+                    finalize=self.model.finalize_backward,
+                    zero_grad=self._needs_zero)
 
     def _try_fast_engine(self, kind: str) -> bool:
         """Engage the toy fast path when the configuration qualifies:
@@ -231,7 +252,7 @@ class Trainer:
         if self._engine is not None:
             self._engine.step(source.contiguous(), targets.contiguous())
             return
-        if not self._distributed:
+        if not self._distributed or getattr(self, "_needs_zero", False):
             self.optimizer.zero_grad(set_to_none=False)
         output = self.model(source)
         loss = self.loss_fn(output, targets)
