@@ -207,7 +207,10 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32, use_mse=True):
 
     noflush = lambda: None  # noqa: E731
     if kind in ("autograd", "autograd-graph"):
-        engine = DDP(model, comm=comm)
+        # capture requires Python hooks (a C++ node post-hook segfaults
+        # hipStreamEndCapture — reducer.py); replay runs no hooks at all
+        engine = DDP(model, comm=comm,
+                     cpp_hooks=False if kind == "autograd-graph" else None)
         opt = FusedSGD(model.parameters(), lr=lr)
         opt.attach_reducer(engine.reducer)
 

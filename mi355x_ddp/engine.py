@@ -213,7 +213,7 @@ class PersistentToyStep(ToyFusedStep):
 
 class GraphedAutogradStep:
     """The GENERIC training step — forward, loss, autograd backward with
-    the ReducerCore hooks, bucket all-reduce, fused SGD — captured in ONE
+    the reducer hooks, bucket all-reduce, fused SGD — captured in ONE
     hipGraph and replayed (torch "whole-network capture").
 
     This is what makes the arbitrary-model path fast on MI355X: eager, the
@@ -222,8 +222,13 @@ class GraphedAutogradStep:
     it pays two tiny input copies plus the graph-replay floor (~10-16 us,
     MI355X_MICROARCH price list). The step is capture-safe BY CONSTRUCTION
     here: gradients are views into the reducer's static flat buckets,
-    FusedSGD updates the static flat pairs, and the C++ hook trampoline
-    launches collectives with stream-event edges capture can record.
+    FusedSGD updates the static flat pairs, and the reducer's bucket
+    collectives are launched with stream-event edges capture records.
+    The wrapped model's reducer must use PYTHON hooks
+    (DDP(..., cpp_hooks=False)): a C++ node post-hook — even a no-op one —
+    segfaults hipStreamEndCapture on this torch/ROCm build (bisected on
+    hardware, tools/capture_bisect2.py). Hooks never execute during
+    replay, so the captured path loses nothing vs the C++ core.
 
     Requirements: static shapes (same batch size every step) and a model
     without data-dependent control flow — the reference workloads (toy

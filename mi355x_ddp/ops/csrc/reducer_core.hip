@@ -1,4 +1,6 @@
 // See reducer_core.h. Host-side C++ (no device code): autograd hook
+#include <cstring>
+#include <cstdlib>
 // trampoline + bucket launch logic for the GPU reducer path.
 #include "reducer_core.h"
 
@@ -23,6 +25,10 @@ struct CoreHook : torch::autograd::FunctionPostHook {
   torch::autograd::variable_list operator()(
       const torch::autograd::variable_list& outputs,
       const torch::autograd::variable_list& /*inputs*/) override {
+    // MI355X_CORE_DEBUG: bisection aid (noop skips the whole body) — used
+    // to isolate the capture-time segfault investigation; not a prod knob.
+    static const char* dbg = getenv("MI355X_CORE_DEBUG");
+    if (dbg && strcmp(dbg, "noop") == 0) return outputs;
     if (auto c = core.lock()) {  // expired core: reducer was dropped — no-op
       c->mark_ready(bucket, index);
     }
@@ -92,9 +98,19 @@ void ReducerCore::detach_hooks() {
 }
 
 void ReducerCore::mark_ready(size_t bucket, size_t index) {
+  static const char* dbg = getenv("MI355X_CORE_DEBUG");
   Bucket& bk = buckets_[bucket];
   const at::Tensor& p = bk.params[index];
   const at::Tensor& view = bk.views[index];
+  if (dbg && strcmp(dbg, "norebind") == 0) {
+    std::lock_guard<std::mutex> lk2(mu_);
+    bk.pending -= 1;
+    if (bk.pending == 0) {
+      bk.ready = true;
+      launch_ready_locked();
+    }
+    return;
+  }
   // Views transport: autograd normally accumulates IN PLACE into our view
   // (p.grad was bound to it at construction). If the engine replaced the
   // grad tensor (first iteration after p.grad=None, or an out-of-place

@@ -32,7 +32,8 @@ class DDP(torch.nn.Module):
                  device_ids=None, output_device=None,
                  bucket_cap_mb: Optional[float] = None,
                  comm=None, grad_views: Optional[bool] = None,
-                 broadcast_buffers: bool = True):
+                 broadcast_buffers: bool = True,
+                 cpp_hooks: Optional[bool] = None):
         # device_ids/output_device: accepted for drop-in compatibility with
         # the stock signature the reference uses (`DDP(model,
         # device_ids=[gpu_id])`, ref multigpu.py:36). The model must already
@@ -63,7 +64,8 @@ class DDP(torch.nn.Module):
         self.comm = comm if comm is not None else create_comm(device)
         self.reducer = Reducer(params, comm=self.comm,
                                bucket_cap_mb=bucket_cap_mb,
-                               grad_views=grad_views)
+                               grad_views=grad_views,
+                               cpp_hooks=cpp_hooks)
         # wrap-time module-state sync: params coalesced per bucket,
         # buffers (e.g. BN running stats) coalesced per dtype (SURVEY §2.4
         # row 2). With broadcast_buffers=True (stock DDP's default) the

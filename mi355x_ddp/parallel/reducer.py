@@ -81,7 +81,8 @@ class Bucket:
 class Reducer:
     def __init__(self, params: List[torch.nn.Parameter], comm=None,
                  bucket_cap_mb: Optional[float] = None,
-                 grad_views: Optional[bool] = None):
+                 grad_views: Optional[bool] = None,
+                 cpp_hooks: Optional[bool] = None):
         params = [p for p in params if p.requires_grad]
         if not params:
             raise ValueError("Reducer needs at least one trainable parameter")
@@ -154,9 +155,15 @@ class Reducer:
         self._core = None
         self._hooks = []
         self._skip_comm = False  # set by DDP.no_sync() during accumulation
+        if cpp_hooks is None:
+            cpp_hooks = os.environ.get("MI355X_CPP_HOOKS", "1") != "0"
+        # cpp_hooks=False is REQUIRED under whole-step hipGraph capture
+        # (GraphedAutogradStep): a C++ node post-hook — even a no-op one —
+        # segfaults hipStreamEndCapture on this torch/ROCm build (bisected
+        # on hardware, profiles/README.md r02). Hooks never execute during
+        # replay, so the captured path loses nothing.
         use_core = (self.grad_views and device.type == "cuda"
-                    and os.environ.get("MI355X_CPP_HOOKS", "1") != "0"
-                    and ops.has_ext())
+                    and cpp_hooks and ops.has_ext())
         core_comm = None
         if use_core and comm is not None:
             from .comm import P2pMeshComm, RcclCommAdapter

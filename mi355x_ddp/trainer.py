@@ -137,7 +137,9 @@ class Trainer:
             # world 1 wraps too: the engine still provides flat buckets and
             # the fused SGD path (there is simply no communicator).
             from .parallel import DDP, FusedSGD
-            self.model = DDP(self.model, bucket_cap_mb=bucket_cap_mb)
+            graphed = engine == "hooks-graph" and self.device.type == "cuda"
+            self.model = DDP(self.model, bucket_cap_mb=bucket_cap_mb,
+                             cpp_hooks=False if graphed else None)
             self._distributed = True
             if isinstance(self.optimizer, FusedSGD):
                 self.optimizer.attach_reducer(self.model.reducer)

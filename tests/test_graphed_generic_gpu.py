@@ -33,8 +33,10 @@ def _batches(n, seed=21):
 
 
 def _build(seed):
+    # capture needs Python hooks (C++ node post-hooks segfault
+    # hipStreamEndCapture on this build — reducer.py); replay runs none
     m = _mlp(seed)
-    eng = DDP(m, comm=None)  # world 1: flat buckets + core hooks, no comm
+    eng = DDP(m, comm=None, cpp_hooks=False)
     opt = FusedSGD(m.parameters(), lr=0.01)
     opt.attach_reducer(eng.reducer)
     return m, eng, opt
@@ -85,8 +87,8 @@ def test_graphed_step_with_world1_rccl_capture():
         data = _batches(5, seed=33)
 
         m_g = _mlp(1)
-        eng_g = DDP(m_g, comm=comm, bucket_cap_mb=0.05)
-        assert eng_g.reducer._core is not None
+        eng_g = DDP(m_g, comm=comm, bucket_cap_mb=0.05, cpp_hooks=False)
+        assert eng_g.reducer._core is None  # Python hooks: capture-safe
         opt_g = FusedSGD(m_g.parameters(), lr=0.01)
         opt_g.attach_reducer(eng_g.reducer)
         gs = GraphedAutogradStep(eng_g, ops.mse_loss, opt_g,
@@ -98,7 +100,7 @@ def test_graphed_step_with_world1_rccl_capture():
             "capture with an in-graph RCCL collective must succeed"
 
         m_e = _mlp(1)
-        eng_e = DDP(m_e, comm=None, bucket_cap_mb=0.05)
+        eng_e = DDP(m_e, comm=None, bucket_cap_mb=0.05, cpp_hooks=False)
         opt_e = FusedSGD(m_e.parameters(), lr=0.01)
         opt_e.attach_reducer(eng_e.reducer)
         ge = GraphedAutogradStep(eng_e, ops.mse_loss, opt_e,
@@ -118,7 +120,7 @@ def test_graphed_resnet50_steps():
     from mi355x_ddp.models import resnet50
     torch.manual_seed(0)
     m = resnet50().to(DEV)
-    eng = DDP(m, comm=None)
+    eng = DDP(m, comm=None, cpp_hooks=False)
     opt = FusedSGD(m.parameters(), lr=1e-4)
     opt.attach_reducer(eng.reducer)
     gs = GraphedAutogradStep(eng, ops.cross_entropy, opt, warmup_steps=2,
