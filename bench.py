@@ -220,7 +220,7 @@ def build_engine(kind, comm, lr, device, dtype=torch.float32, use_mse=True):
             from mi355x_ddp.engine import GraphedAutogradStep
             gs = GraphedAutogradStep(engine, loss_fn, opt,
                                      finalize=engine.finalize_backward)
-            return gs.step, noflush, None
+            return gs.step, gs.flush, gs  # shard-bound protocol supported
 
         def step(x, t):
             loss = loss_fn(engine(x), t)
@@ -430,6 +430,7 @@ def main():
                            args.engine if engine_obj is None else
                            {"PersistentToyStep": "persistent",
                             "GraphedToyStep": "graph",
+                            "GraphedAutogradStep": "autograd-graph",
                             "ToyFusedStep": "fused"}[
                                type(engine_obj).__name__]),
                 "loss": args.loss,

@@ -230,6 +230,16 @@ class GraphedAutogradStep:
     hardware, tools/capture_bisect2.py). Hooks never execute during
     replay, so the captured path loses nothing vs the C++ core.
 
+    Deferred multi-step replay: besides the single-step `step(x, t)` API,
+    the engine speaks the shard-bound protocol (`bind_shard`/`step_shard`/
+    `flush`, same contract as PersistentToyStep): runs of consecutive
+    batch indices execute as replays of MULTI-STEP graphs — a G-step graph
+    amortizes the replay floor and the input copies over G steps, leaving
+    the in-graph kernel-boundary cost (~1.2-1.5 us per dependent kernel)
+    as the bound. Runs decompose greedily over `chunk_sizes` graphs, each
+    captured once on first use; warmup covers the steady-state sizes so
+    captures stay out of timed regions.
+
     Requirements: static shapes (same batch size every step) and a model
     without data-dependent control flow — the reference workloads (toy
     Linear, ResNet-50) qualify. Capture failure falls back to eager with a
@@ -237,7 +247,8 @@ class GraphedAutogradStep:
     """
 
     def __init__(self, model, loss_fn, optimizer, finalize=None,
-                 warmup_steps: int = 3, zero_grad: bool = False):
+                 warmup_steps: int = 3, zero_grad: bool = False,
+                 chunk_sizes=(64, 8, 1)):
         self.model = model            # DDP-wrapped or bare
         self.loss_fn = loss_fn
         self.optimizer = optimizer
@@ -248,9 +259,18 @@ class GraphedAutogradStep:
         # the captured step, set_to_none=False so the grad buffers stay
         # static (a capture requirement, and it keeps the bucket views).
         self._zero = zero_grad
-        self._graph = None
-        self._x = self._t = None
-        self.loss = None              # captured loss tensor (replay target)
+        self.chunk_sizes = tuple(sorted(set(chunk_sizes) | {1},
+                                        reverse=True))
+        self.max_defer = self.chunk_sizes[0]
+        self._graphs = {}             # G -> (graph, xbuf, tbuf)
+        self._broken = False          # capture failed: eager forever
+        self._warmed = False
+        self._bshape = None           # (x batch shape, t batch shape)
+        self.loss = None              # last captured loss tensor
+        # shard-bound state
+        self._shard_x = self._shard_t = None
+        self._sbatch = 0
+        self._s_start = self._s_next = 0
 
     def _eager_step(self, x, t):
         if self._zero:
@@ -261,49 +281,116 @@ class GraphedAutogradStep:
         self.optimizer.step()
         return loss
 
-    def _capture(self, x, t):
-        self._x = x.clone()
-        self._t = t.clone()
-        # torch's capture recipe: warm up on a side stream first (allocator
-        # + autograd engine touch lazily-initialized state off-capture)
+    def _warm(self, x, t):
+        """torch's capture recipe: a few eager steps on a side stream
+        before the first capture (allocator + autograd engine initialize
+        lazily-created state off-capture). These steps train for real —
+        on the first batch, like any warmup."""
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s):
             for _ in range(self.warmup_steps):
-                self._eager_step(self._x, self._t)
+                self._eager_step(x, t)
         torch.cuda.current_stream().wait_stream(s)
         torch.cuda.synchronize()
+        self._warmed = True
+
+    def _get_graph(self, G):
+        """The G-step graph (captured on first use). Returns None when
+        capture is broken; xbuf/tbuf hold G consecutive batches."""
+        got = self._graphs.get(G)
+        if got is not None or self._broken:
+            return got
+        B = self._bshape[0][0]
+        xbuf = torch.empty((G * B,) + self._bshape[0][1:],
+                           dtype=self._bdtype[0], device=self._bdev)
+        tbuf = torch.empty((G * B,) + self._bshape[1][1:],
+                           dtype=self._bdtype[1], device=self._bdev)
         try:
             g = torch.cuda.CUDAGraph()
             with torch.cuda.graph(g):
-                self.loss = self._eager_step(self._x, self._t)
-            self._graph = g
+                for i in range(G):
+                    self.loss = self._eager_step(xbuf[i * B:(i + 1) * B],
+                                                 tbuf[i * B:(i + 1) * B])
+            self._graphs[G] = (g, xbuf, tbuf)
+            return self._graphs[G]
         except Exception as e:
             import warnings
             warnings.warn(
                 f"[mi355x_ddp] whole-step hipGraph capture failed ({e!r}); "
                 "GraphedAutogradStep running eager (correct, slower)",
                 RuntimeWarning, stacklevel=2)
-            self._graph = False
+            self._broken = True
             torch.cuda.synchronize()
+            return None
+
+    def _note_shapes(self, x, t):
+        if self._bshape is None:
+            self._bshape = (x.shape, t.shape)
+            self._bdtype = (x.dtype, t.dtype)
+            self._bdev = x.device
 
     def step(self, x, t):
-        if self._graph is None:
-            self._capture(x, t)
-            if self._graph:
-                self._graph.replay()  # capture records, replay executes:
-                # this call still performs exactly one (post-warmup) step
-            else:
-                self._eager_step(x, t)
+        """Single-step API (Trainer): one G=1 graph replay per call."""
+        self.flush()
+        self._note_shapes(x, t)
+        if not self._broken and x.shape != self._bshape[0]:
+            self._eager_step(x, t)  # ragged final batch: run it eager
             return
-        if self._graph is False or x.shape != self._x.shape:
-            # eager fallback; shape mismatch = a ragged final batch — run
-            # it eager rather than replaying a graph of the wrong shape
+        if not self._warmed:
+            self._warm(x, t)
+        got = self._get_graph(1)
+        if got is None:
             self._eager_step(x, t)
             return
-        self._x.copy_(x, non_blocking=True)
-        self._t.copy_(t, non_blocking=True)
-        self._graph.replay()
+        g, xbuf, tbuf = got
+        xbuf.copy_(x.reshape(xbuf.shape), non_blocking=True)
+        tbuf.copy_(t.reshape(tbuf.shape), non_blocking=True)
+        g.replay()
+
+    # -- shard-bound protocol (same contract as PersistentToyStep) -------
+    def bind_shard(self, xs, ts, batch):
+        self.flush()
+        assert xs.is_cuda and xs.is_contiguous() and ts.is_contiguous()
+        self._shard_x, self._shard_t, self._sbatch = xs, ts, batch
+        self._s_start = self._s_next = 0
+        self._note_shapes(xs[:batch], ts[:batch])
+
+    def step_shard(self, i):
+        if i == self._s_next:
+            self._s_next += 1
+            if self._s_next - self._s_start >= self.max_defer:
+                self.flush()
+            return
+        self.flush()
+        self._s_start, self._s_next = i, i + 1
+
+    def flush(self):
+        if self._s_next <= self._s_start:
+            return
+        lo, n = self._s_start, self._s_next - self._s_start
+        self._s_start = self._s_next
+        B = self._sbatch
+        if not self._warmed:
+            self._warm(self._shard_x[lo * B:(lo + 1) * B],
+                       self._shard_t[lo * B:(lo + 1) * B])
+        while n:
+            for G in self.chunk_sizes:
+                if G <= n:
+                    break
+            s, e = lo * B, (lo + G) * B
+            got = self._get_graph(G)
+            if got is None:  # capture broken: eager the rest of the run
+                for i in range(lo, lo + n):
+                    self._eager_step(self._shard_x[i * B:(i + 1) * B],
+                                     self._shard_t[i * B:(i + 1) * B])
+                return
+            g, xbuf, tbuf = got
+            xbuf.copy_(self._shard_x[s:e], non_blocking=True)
+            tbuf.copy_(self._shard_t[s:e], non_blocking=True)
+            g.replay()
+            lo += G
+            n -= G
 
 
 class GraphedToyStep(ToyFusedStep):

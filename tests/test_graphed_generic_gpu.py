@@ -53,7 +53,7 @@ def test_graphed_step_matches_eager_bitwise():
     for x, t in data:
         gs.step(x, t)
     torch.cuda.synchronize()
-    assert gs._graph not in (None, False), "capture must succeed for MLP"
+    assert not gs._broken and 1 in gs._graphs, "capture must succeed for MLP"
 
     # eager arm replicating the capture call's schedule: W warmup steps on
     # batch0, then the captured step on batch0, then batches 1..5
@@ -96,7 +96,7 @@ def test_graphed_step_with_world1_rccl_capture():
         for x, t in data:
             gs.step(x, t)
         torch.cuda.synchronize()
-        assert gs._graph not in (None, False), \
+        assert not gs._broken and 1 in gs._graphs, \
             "capture with an in-graph RCCL collective must succeed"
 
         m_e = _mlp(1)
@@ -161,7 +161,7 @@ def test_trainer_hooks_graph_engine(tmp_path):
         assert tr._engine is not None, "hooks-graph must install the engine"
         tr.train(3)
         torch.cuda.synchronize()
-        assert tr._engine._graph not in (None, False), \
+        assert not tr._engine._broken and tr._engine._graphs, \
             "toy step must capture"
         got = [p.detach() for p in tr._unwrapped().parameters()]
         assert all(torch.isfinite(p).all() for p in got)
@@ -172,3 +172,41 @@ def test_trainer_hooks_graph_engine(tmp_path):
         toy_model(20, 1).load_state_dict(sd)
     finally:
         dist.destroy_process_group()
+
+
+def test_graphed_shard_bound_matches_eager_bitwise():
+    # the multi-step chunked-replay path (bind_shard/step_shard/flush):
+    # S=13 decomposes as one 8-step graph + five 1-step graphs
+    S, B = 13, 16
+    g = torch.Generator().manual_seed(55)
+    xs = torch.randn(S * B, 64, generator=g).to(DEV)
+    ts = torch.randn(S * B, 10, generator=g).to(DEV)
+
+    m_g, eng_g, opt_g = _build(6)
+    gs = GraphedAutogradStep(eng_g, ops.mse_loss, opt_g,
+                             finalize=eng_g.finalize_backward,
+                             chunk_sizes=(8, 1))
+    gs.bind_shard(xs, ts, B)
+    for i in range(S):
+        gs.step_shard(i)
+    gs.flush()
+    torch.cuda.synchronize()
+    assert not gs._broken and 8 in gs._graphs and 1 in gs._graphs
+
+    # eager arm: warmup trains warmup_steps times on batch 0, then each
+    # batch in order
+    m_e, eng_e, opt_e = _build(6)
+
+    def eager(lo, hi):
+        loss = ops.mse_loss(eng_e(xs[lo:hi]), ts[lo:hi])
+        loss.backward()
+        eng_e.finalize_backward()
+        opt_e.step()
+
+    for _ in range(gs.warmup_steps):
+        eager(0, B)
+    for i in range(S):
+        eager(i * B, (i + 1) * B)
+    torch.cuda.synchronize()
+    for a, b in zip(m_g.parameters(), m_e.parameters()):
+        assert torch.equal(a, b), (a - b).abs().max()
