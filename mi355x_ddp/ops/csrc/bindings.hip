@@ -100,7 +100,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
       .def("finalize", &mi355x::ReducerCore::finalize,
            py::call_guard<py::gil_scoped_release>())
       .def("set_skip_comm", &mi355x::ReducerCore::set_skip_comm)
-      .def_property_readonly("steps", &mi355x::ReducerCore::steps);
+      .def_property_readonly("steps", &mi355x::ReducerCore::steps)
+      .def_property_readonly("unfenced", &mi355x::ReducerCore::unfenced);
 
   py::class_<mi355x::RcclComm>(m, "RcclComm")
       .def(py::init<const std::string&, int, int, int>(), py::arg("unique_id"),

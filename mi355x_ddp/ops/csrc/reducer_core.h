@@ -61,6 +61,12 @@ class ReducerCore : public std::enable_shared_from_this<ReducerCore> {
   void set_skip_comm(bool v) { skip_comm_ = v; }
   bool skip_comm() const { return skip_comm_; }
   int64_t steps() const { return steps_; }
+  // True while bucket collectives have been launched on the comm stream
+  // but finalize() has not yet fenced the compute stream on them —
+  // reading flat_grad in that window is the reducer's one real race
+  // (SURVEY §5.2). Consumers (FusedSGD) check this under
+  // MI355X_DEBUG_SYNC=1.
+  bool unfenced() const { return unfenced_; }
 
  private:
   struct Bucket {
@@ -76,6 +82,7 @@ class ReducerCore : public std::enable_shared_from_this<ReducerCore> {
   std::vector<Bucket> buckets_;
   RcclComm* comm_;  // non-owning; the Python adapter keeps it alive
   bool skip_comm_ = false;
+  bool unfenced_ = false;
   int next_launch_ = 0;
   int64_t steps_ = 0;
   std::mutex mu_;

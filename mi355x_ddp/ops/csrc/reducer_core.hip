@@ -136,6 +136,7 @@ void ReducerCore::launch_ready_locked() {
          buckets_[next_launch_].ready) {
     if (comm_ != nullptr && !skip_comm_) {
       comm_->all_reduce_avg(buckets_[next_launch_].flat_grad);
+      unfenced_ = true;
     }
     ++next_launch_;
   }
@@ -153,6 +154,7 @@ void ReducerCore::finalize() {
   if (comm_ != nullptr && !skip_comm_) {
     comm_->join_compute();
   }
+  unfenced_ = false;
   for (auto& bk : buckets_) {
     bk.pending = (int)bk.params.size();
     bk.ready = false;

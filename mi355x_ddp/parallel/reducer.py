@@ -155,6 +155,7 @@ class Reducer:
         self._core = None
         self._hooks = []
         self._skip_comm = False  # set by DDP.no_sync() during accumulation
+        self._unfenced = False   # collectives launched but not yet fenced
         if cpp_hooks is None:
             cpp_hooks = os.environ.get("MI355X_CPP_HOOKS", "1") != "0"
         # cpp_hooks=False is REQUIRED under whole-step hipGraph capture
@@ -228,6 +229,16 @@ class Reducer:
                 self._launch_ready_in_order()
         return hook
 
+    @property
+    def unfenced(self) -> bool:
+        """True while bucket collectives are launched but not yet fenced
+        by finalize() — reading flat_grad in this window is the reducer's
+        one real race. Checked by FusedSGD under
+        MI355X_DEBUG_SYNC=1."""
+        if self._core is not None:
+            return self._core.unfenced
+        return self._unfenced
+
     def _launch_ready_in_order(self) -> None:
         while (self._next_launch < len(self.buckets)
                and self.buckets[self._next_launch].ready):
@@ -238,6 +249,7 @@ class Reducer:
                 self._flatten_bucket(b)
             if self.comm is not None and not self.skip_comm:
                 self.comm.all_reduce_avg(b.flat_grad)
+                self._unfenced = True
             self._next_launch += 1
 
     def _flatten_bucket(self, b: Bucket) -> None:
@@ -292,6 +304,7 @@ class Reducer:
         self._next_launch = len(self.buckets)
         if self.comm is not None and not self.skip_comm:
             self.comm.join_compute()
+        self._unfenced = False
         for b in self.buckets:
             b.pending = len(b.params)
             b.ready = False

@@ -36,6 +36,7 @@ class FusedSGD(torch.optim.Optimizer):
                              "param_group (flat buckets take one lr)")
         self._flat_pairs = reducer.flat_pairs()
         self._bucketed = {id(p) for b in reducer.buckets for p in b.params}
+        self._reducer = reducer
 
     @torch.no_grad()
     def step(self, closure=None):
@@ -43,6 +44,20 @@ class FusedSGD(torch.optim.Optimizer):
         if closure is not None:
             with torch.enable_grad():
                 loss = closure()
+        # Stream-race assertion (MI355X_DEBUG_SYNC=1): the
+        # fused step reads flat_grad on the compute stream — if bucket
+        # collectives were launched but finalize() never fenced them, that
+        # read races the comm stream. Catch the protocol violation loudly
+        # instead of training on half-reduced gradients.
+        import os
+        if (os.environ.get("MI355X_DEBUG_SYNC") == "1"
+                and getattr(self, "_reducer", None) is not None
+                and self._reducer.unfenced):
+            raise RuntimeError(
+                "FusedSGD.step() called while bucket all-reduces are "
+                "in flight and unfenced — call finalize_backward() (or "
+                "Reducer.finalize()) between loss.backward() and "
+                "optimizer.step()")
         for group in self.param_groups:
             lr = group["lr"]
             if self._flat_pairs is not None:

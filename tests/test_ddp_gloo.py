@@ -362,3 +362,43 @@ def test_unused_parameters_do_not_hang(tmp_path):
         assert p.exitcode == 0
     got = torch.load(out_path, weights_only=True)
     assert all(torch.isfinite(p).all() for p in got)
+
+
+def _race_worker(rank, port, out_dir):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["MI355X_DEBUG_SYNC"] = "1"
+    torch.distributed.init_process_group("gloo", rank=rank, world_size=WORLD)
+    try:
+        model = _make_model(0)
+        engine = DDP(model)
+        opt = FusedSGD(model.parameters(), lr=LR)
+        opt.attach_reducer(engine.reducer)
+        x, t = torch.randn(8, 20), torch.randn(8, 1)
+        loss = ops.mse_loss(engine(x), t)
+        loss.backward()
+        # DELIBERATE ordering corruption: step() without finalize —
+        # collectives launched by the hooks are still unfenced
+        caught = False
+        try:
+            opt.step()
+        except RuntimeError as e:
+            caught = "unfenced" in str(e)
+        # recover properly and verify the good ordering passes
+        engine.finalize_backward()
+        opt.step()
+        with open(os.path.join(out_dir, f"race{rank}.ok"), "w") as f:
+            f.write("caught" if caught else "missed")
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def test_debug_sync_catches_step_before_finalize(tmp_path):
+    """the reducer's one real race is reading flat_grad while
+    bucket collectives are unfenced. MI355X_DEBUG_SYNC=1 must catch the
+    corrupted ordering (step before finalize) and pass the correct one."""
+    mp.spawn(_race_worker, args=(_free_port(), str(tmp_path)),
+             nprocs=WORLD, join=True)
+    for r in range(WORLD):
+        assert (tmp_path / f"race{r}.ok").read_text() == "caught"

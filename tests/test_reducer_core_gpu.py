@@ -188,3 +188,32 @@ def test_core_world1_rccl_comm(tmp_path):
             assert torch.equal(a, b)
     finally:
         dist.destroy_process_group()
+
+
+def test_core_debug_sync_catches_unfenced_step(monkeypatch):
+    # GPU twin of the gloo stream-race test: the C++ core launches the
+    # bucket collective from its hook; stepping before finalize() must be
+    # caught under MI355X_DEBUG_SYNC=1
+    import torch.distributed as dist
+    monkeypatch.setenv("MI355X_DEBUG_SYNC", "1")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29784")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        from mi355x_ddp.parallel.comm import RcclCommAdapter
+        comm = RcclCommAdapter(torch.device(DEV))
+        m = _mlp(8)
+        red = Reducer(list(m.parameters()), comm=comm, grad_views=True)
+        assert red._core is not None
+        opt = FusedSGD(m.parameters(), lr=0.01)
+        opt.attach_reducer(red)
+        x = torch.randn(16, 64, device=DEV)
+        t = torch.randn(16, 10, device=DEV)
+        torch.nn.functional.mse_loss(m(x), t).backward()
+        with pytest.raises(RuntimeError, match="unfenced"):
+            opt.step()
+        red.finalize()
+        opt.step()  # correct ordering passes
+        torch.cuda.synchronize()
+    finally:
+        dist.destroy_process_group()
