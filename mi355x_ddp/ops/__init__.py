@@ -69,6 +69,13 @@ def linear(x: torch.Tensor, w: torch.Tensor, b: Optional[torch.Tensor] = None):
     in C++ (csrc/autograd_ops.hip) so the backward chain never re-enters
     Python — the dominant cost of the round-1 generic path."""
     if x.is_cuda:
+        if torch.is_autocast_enabled("cuda"):
+            # nn.Linear's autocast policy: run the matmul in the autocast
+            # dtype (mixed-precision resnet/bf16 stage). Without this, a
+            # bf16 activation meeting an fp32 weight is a dtype error.
+            dt = torch.get_autocast_dtype("cuda")
+            x, w = x.to(dt), w.to(dt)
+            b = b.to(dt) if b is not None else None
         if x.dim() != 2:
             # nn.Linear semantics for arbitrary leading dims: flatten to
             # the kernels' 2-D contract, restore after (reshape is
@@ -95,6 +102,11 @@ def cross_entropy(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
     == 0; SURVEY §2.1 'Degenerate loss') — both paths reproduce torch's
     exact semantics for it.
     """
+    if output.is_cuda and torch.is_autocast_enabled("cuda") \
+            and output.dtype != torch.float32:
+        # torch's autocast policy runs losses in fp32; match it so the
+        # mixed-precision (bf16-autocast) stage keeps fp32 loss numerics
+        output = output.float()
     if not target.is_floating_point():
         if (target < 0).any():
             # torch maps negative indices to ignore_index semantics
@@ -117,6 +129,9 @@ def cross_entropy(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
 
 def mse_loss(output: torch.Tensor, target: torch.Tensor) -> torch.Tensor:
     if output.is_cuda:
+        if torch.is_autocast_enabled("cuda") \
+                and output.dtype != torch.float32:
+            output = output.float()  # autocast loss policy: fp32
         return ext().mse_autograd(output, target.to(output.dtype))
     return F.mse_loss(output, target.to(output.dtype))
 

@@ -1273,13 +1273,15 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
                       int mworld = 1, float minv_world = 1.f,
                       unsigned long long seq0 = 0,
                       unsigned int* __restrict__ mesh_err = nullptr) {
-  static_assert(B_ == 32 && K_ <= 32, "bf16 wide path: B=32, K<=32");
-  constexpr int MT = 2;                // fwd 16-row tiles
+  static_assert((B_ == 32 || B_ == 64) && K_ <= 32,
+                "bf16 wide path: B in {32, 64}, K<=32");
+  constexpr int MT = B_ / 16;          // fwd 16-row tiles
+  constexpr int BT = B_ / 32;          // bwd contraction chunks (MFMA K=32)
   constexpr int KT = (K_ + 15) / 16;   // bwd 16-col tiles
   const int lane = threadIdx.x;
   const int r = lane & 15, q = lane >> 4;
   __shared__ float ws[33];
-  __shared__ float dy_s[32];
+  __shared__ float dy_s[B_];
 
   if (lane < K_) ws[lane] = ldf(&param[w_off + lane]);
   if (lane == K_) ws[32] = ldf(&param[b_off]);
@@ -1300,9 +1302,9 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
   // adopt-copies, no issue at the chain head — the two failure modes of
   // the earlier pipelining attempts, kept in git history).
   struct BSet {
-    bf16x8 fa[MT];     // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
-    bf16x8 bb[KT];     // bwd B: X[q*8+j][tk*16+r] (raw)
-    float tR[MT][4];   // targets for rows tm*16+q*4+i
+    bf16x8 fa[MT];      // fwd A: X[tm*16+r][q*8+j] (raw, clamped)
+    bf16x8 bb[BT][KT];  // bwd B: X[bt*32 + q*8+j][tk*16+r] (raw)
+    float tR[MT][4];    // targets for rows tm*16+q*4+i
   };
   BSet setA, setB;
 
@@ -1320,15 +1322,18 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
       }
     }
 #pragma unroll
-    for (int tk = 0; tk < KT; ++tk) {
-      const int k = tk * 16 + r;
-      const int kc = (k < K_) ? k : K_ - 1;
+    for (int bt = 0; bt < BT; ++bt)
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int i = q * 8 + j;  // < 32 == B_
-        R.bb[tk][j] = *reinterpret_cast<const __bf16*>(&Xs[(size_t)i * K_ + kc]);
+      for (int tk = 0; tk < KT; ++tk) {
+        const int k = tk * 16 + r;
+        const int kc = (k < K_) ? k : K_ - 1;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int i = bt * 32 + q * 8 + j;  // batch row of this chunk
+          R.bb[bt][tk][j] =
+              *reinterpret_cast<const __bf16*>(&Xs[(size_t)i * K_ + kc]);
+        }
       }
-    }
 #pragma unroll
     for (int tm = 0; tm < MT; ++tm)
 #pragma unroll
@@ -1341,7 +1346,7 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
   float loss_last = 0.f;
   auto body = [&](BSet& R, int s) {
     const bf16x8* cfa = R.fa;
-    const bf16x8* cbb = R.bb;
+    const auto& cbb = R.bb;
     const auto& ctR = R.tR;
     // forward B-operand: w as bf16 (stored values round-trip bf16 exactly),
     // zero-padded for k >= K_ so raw garbage in A contributes nothing
@@ -1380,26 +1385,33 @@ k_toy_multistep_bf16w(const __hip_bfloat16* __restrict__ X,
     }
     __syncthreads();  // dy_s visible to all lanes
 
-    // backward A-operand: dY rounded to bf16 (j-th element = dy[q*8+j])
-    bf16x8 a8{};
+    // backward A-operands: dY rounded to bf16, one 32-row chunk per BT
+    // (j-th element of chunk bt = dy[bt*32 + q*8+j]); chunks accumulate
+    // into the same f32 accumulator — the full B_ contraction.
+    bf16x8 a8[BT];
 #pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const float v = dy_s[q * 8 + j];
-      a8[j] = (r == 0) ? (__bf16)v : (__bf16)0.f;
-    }
+    for (int bt = 0; bt < BT; ++bt)
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float v = dy_s[bt * 32 + q * 8 + j];
+        a8[bt][j] = (r == 0) ? (__bf16)v : (__bf16)0.f;
+      }
     f32x4 gacc[KT];
     static_assert(K_ % 16 != 0, "free db column requires K_ % 16 != 0");
 #pragma unroll
     for (int tk = 0; tk < KT; ++tk) {
-      // db rides output column K_: constant-1 B operand on that lane
-      bf16x8 bb = cbb[tk];
-      if (tk * 16 + r == K_) {
-#pragma unroll
-        for (int j = 0; j < 8; ++j) bb[j] = (__bf16)1.f;
-      }
       gacc[tk] = {0.f, 0.f, 0.f, 0.f};
-      gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a8, bb,
-                                                         gacc[tk], 0, 0, 0);
+#pragma unroll
+      for (int bt = 0; bt < BT; ++bt) {
+        // db rides output column K_: constant-1 B operand on that lane
+        bf16x8 bb = cbb[bt][tk];
+        if (tk * 16 + r == K_) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) bb[j] = (__bf16)1.f;
+        }
+        gacc[tk] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a8[bt], bb,
+                                                           gacc[tk], 0, 0, 0);
+      }
     }
 
     if (s + 2 < S) prefetch(R, s + 2);  // this set's regs dead from here
@@ -1657,15 +1669,19 @@ static void launch_toy_multistep(const torch::Tensor& x, const torch::Tensor& t,
     }
     return;
   }
-  if (B == 64 && K == 20 && !std::is_same<T, __hip_bfloat16>::value) {
-    // batch-64 variant of the fast path (bf16 wide kernel is 32-only;
-    // bf16 at B=64 takes the generic rolled kernel below)
-    if constexpr (!std::is_same<T, __hip_bfloat16>::value) {
+  if (B == 64 && K == 20) {  // batch-64 variant of the fast path
+    if constexpr (std::is_same<T, __hip_bfloat16>::value) {
+      // wide bf16 MFMA, backward contraction as two chained 16x16x32
+      // chunks (close the fast-path matrix)
+      hipLaunchKernelGGL((k_toy_multistep_bf16w<64, 20>), dim3(1), dim3(64),
+                         0, cur_stream(), xp, tp, pp, lossp, S,
+                         use_mse ? 1 : 0, w_off, b_off, lr);
+    } else {
       hipLaunchKernelGGL((k_toy_multistep_spec<T, 64, 20>), dim3(1), dim3(64),
                          0, cur_stream(), xp, tp, pp, lossp, S,
                          use_mse ? 1 : 0, w_off, b_off, lr);
-      return;
     }
+    return;
   }
   auto go = [&](auto mt, auto kt) {
     hipLaunchKernelGGL((k_toy_multistep<T, decltype(mt)::value, decltype(kt)::value>),

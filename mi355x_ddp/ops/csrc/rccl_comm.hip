@@ -1,4 +1,6 @@
 // See rccl_comm.h. Native HIP + RCCL; no CUDA compatibility paths.
+#include <cstring>
+#include <cstdlib>
 #include "rccl_comm.h"
 
 #include <ATen/hip/HIPContext.h>
@@ -49,9 +51,17 @@ RcclComm::RcclComm(const std::string& unique_id, int rank, int world, int device
   HIP_OK(hipSetDevice(device));
   // Dedicated, high-priority comm stream: bucket all-reduces launched here
   // overlap with backward compute on the torch stream.
-  int least = 0, greatest = 0;
-  HIP_OK(hipDeviceGetStreamPriorityRange(&least, &greatest));
-  HIP_OK(hipStreamCreateWithPriority(&comm_stream_, hipStreamNonBlocking, greatest));
+  // MI355X_COMM_PRIO=0 drops the priority (diagnostic: high-priority
+  // streams were a suspect in the MIOpen find-time solution regression).
+  const char* prio_env = getenv("MI355X_COMM_PRIO");
+  if (prio_env && strcmp(prio_env, "0") == 0) {
+    HIP_OK(hipStreamCreateWithFlags(&comm_stream_, hipStreamNonBlocking));
+  } else {
+    int least = 0, greatest = 0;
+    HIP_OK(hipDeviceGetStreamPriorityRange(&least, &greatest));
+    HIP_OK(hipStreamCreateWithPriority(&comm_stream_, hipStreamNonBlocking,
+                                       greatest));
+  }
   HIP_OK(hipEventCreateWithFlags(&ready_ev_, hipEventDisableTiming));
   HIP_OK(hipEventCreateWithFlags(&done_ev_, hipEventDisableTiming));
   NCCL_OK(ncclCommInitRank(&comm_, world, id, rank));

@@ -352,3 +352,47 @@ def test_spec_multistep_batch64_bitwise():
     torch.cuda.synchronize()
     assert torch.equal(m.weight.detach(), m_ref.weight.detach())
     assert torch.equal(m.bias.detach(), m_ref.bias.detach())
+
+
+def test_bf16_multistep_batch64():
+    # the batch-64 bf16 wide-MFMA instantiation
+    # (backward contraction = two chained 16x16x32 chunks).
+    # (a) one S=10 launch == ten S=1 launches of the SAME kernel, bitwise;
+    # (b) close to the f32-MFMA fused single-step path (bf16 accuracy).
+    from mi355x_ddp.engine import PersistentToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    g = torch.Generator().manual_seed(23)
+    Xf = torch.rand(10 * 64, 20, generator=g).to(DEV).bfloat16()
+    Tf = torch.rand(10 * 64, 1, generator=g).to(DEV).bfloat16()
+
+    def persistent(step_flush):
+        torch.manual_seed(6)
+        m = toy_model(20, 1).to(DEV).bfloat16()
+        eng = PersistentToyStep(m, comm=None, lr=0.04, use_mse=True)
+        eng.bind_shard(Xf, Tf, 64)
+        for s in range(10):
+            eng.step_shard(s)
+            if step_flush:
+                eng.flush()  # forces S=1 launches of the same kernel
+        eng.flush()
+        torch.cuda.synchronize()
+        return (m.weight.detach().float().cpu(),
+                m.bias.detach().float().cpu())
+
+    w_multi, b_multi = persistent(step_flush=False)
+    w_single, b_single = persistent(step_flush=True)
+    assert torch.equal(w_multi, w_single), (w_multi - w_single).abs().max()
+    assert torch.equal(b_multi, b_single)
+
+    torch.manual_seed(6)
+    m_ref = toy_model(20, 1).to(DEV).bfloat16()
+    eager = ToyFusedStep(m_ref, comm=None, lr=0.04, use_mse=True)
+    for s in range(10):
+        eager.step(Xf[s * 64:(s + 1) * 64].contiguous(),
+                   Tf[s * 64:(s + 1) * 64].contiguous())
+    torch.cuda.synchronize()
+    w_ref = m_ref.weight.detach().float().cpu()
+    b_ref = m_ref.bias.detach().float().cpu()
+    assert torch.allclose(w_multi, w_ref, atol=5e-3, rtol=5e-2), \
+        (w_multi - w_ref).abs().max()
+    assert torch.allclose(b_multi, b_ref, atol=5e-3, rtol=5e-2)

@@ -39,9 +39,11 @@ def overlap_from_trace(path):
                                                       "Kernel")]
     rccl, comp = [], []
     for e in evs:
-        (rccl if "nccl" in e.get("name", "").lower()
-         or "rccl" in e.get("name", "").lower() else comp).append(
-            (e["ts"], e["ts"] + e["dur"]))
+        n = e.get("name", "").lower()
+        # RCCL device kernels: multi-rank "ncclDevKernel_*"; world-1
+        # reductions dispatch as "oneRankReduce<...>" (measured on-box)
+        is_rccl = "nccl" in n or "rccl" in n or "onerankreduce" in n
+        (rccl if is_rccl else comp).append((e["ts"], e["ts"] + e["dur"]))
     comp.sort()
     # merge compute intervals
     merged = []
@@ -88,10 +90,21 @@ def main():
         dist.init_process_group("gloo", rank=0, world_size=1)
     torch.cuda.set_device(local)
     device = torch.device("cuda", local)
-    comm = RcclCommAdapter(device)
 
     torch.manual_seed(0)
     model = resnet50().to(device)
+    # MIOpen find BEFORE RCCL init: an initialized RCCL communicator
+    # during the first conv finds makes MIOpen pick ~2.5x slower conv
+    # solutions (measured: 44.3 vs 17.5 ms/step, tools/comm_order_debug).
+    # Warm every fwd+bwd conv shape first, then bring up the comm.
+    xw = torch.rand(args.batch, 3, 224, 224, device=device)
+    tw = torch.rand(args.batch, 1000, device=device)
+    for _ in range(2):
+        ops.cross_entropy(model(xw), tw).backward()
+    for p in model.parameters():
+        p.grad = None
+    torch.cuda.synchronize()
+    comm = RcclCommAdapter(device)
     params = list(model.parameters())
     red = Reducer(params, comm=comm, bucket_cap_mb=args.cap_mb)
     red.broadcast_params(0)
