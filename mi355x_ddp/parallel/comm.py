@@ -65,7 +65,13 @@ def ddp_setup(rank: Optional[int] = None, world_size: Optional[int] = None,
     else:
         dist.init_process_group(backend)
     if torch.cuda.is_available():
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank or 0)))
+        if os.environ.get("MI355X_FORCE_DEV0") == "1":
+            # test-only: rehearse a multi-rank world on ONE device (IPC
+            # time-sharing) — used by the 8-GPU pre-flight on 1-GPU boxes
+            torch.cuda.set_device(0)
+        else:
+            torch.cuda.set_device(int(os.environ.get("LOCAL_RANK",
+                                                     rank or 0)))
 
 
 class GlooComm:
@@ -256,6 +262,8 @@ def _this_rank_device() -> torch.device:
     for a user-initialized group our ddp_setup (which calls
     torch.cuda.set_device) may not have run, and cuda:0 from every local
     rank would collide in NCCL."""
+    if os.environ.get("MI355X_FORCE_DEV0") == "1":
+        return torch.device("cuda", 0)
     if "LOCAL_RANK" in os.environ:
         return torch.device("cuda", int(os.environ["LOCAL_RANK"]))
     return torch.device("cuda", torch.cuda.current_device())

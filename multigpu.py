@@ -55,7 +55,10 @@ def main(rank: int, world_size: int, total_epochs: int, save_every: int):
 if __name__ == "__main__":
     total_epochs = int(sys.argv[1])
     save_every = int(sys.argv[2])
-    if torch.cuda.is_available():
+    if os.environ.get("MI355X_FORCE_DEV0") == "1":
+        # pre-flight rehearsal: N ranks time-sharing one device via IPC
+        world_size = int(os.environ.get("MI355X_WORLD", 2))
+    elif torch.cuda.is_available():
         world_size = torch.cuda.device_count()
     else:
         world_size = int(os.environ.get("MI355X_WORLD", 2))

@@ -78,7 +78,10 @@ if __name__ == "__main__":
     total_epochs = int(sys.argv[1]) if len(sys.argv) > 1 else 3
     dataset_size = int(os.environ.get("MI355X_PROFILE_DATASET", 2048))
     batch_size = int(os.environ.get("MI355X_PROFILE_BATCH", 32))
-    if torch.cuda.is_available():
+    if os.environ.get("MI355X_FORCE_DEV0") == "1":
+        # pre-flight rehearsal: N ranks time-sharing one device via IPC
+        world_size = int(os.environ.get("MI355X_WORLD", 2))
+    elif torch.cuda.is_available():
         world_size = torch.cuda.device_count()
     else:
         world_size = int(os.environ.get("MI355X_WORLD", 2))

@@ -169,3 +169,7 @@ def test_profile_script_tiny(tmp_path):
     traces = list(trace_dir.glob("*.json")) + list(trace_dir.glob("*.json.gz")) \
         + list(trace_dir.glob("*.pt.trace.json*"))
     assert traces, list(trace_dir.iterdir()) if trace_dir.exists() else "no dir"
+    # beyond-parity: the per-rank kernel summary artifacts exist
+    stats = list(trace_dir.glob("kernel_stats_rank*.json"))
+    assert len(stats) == 2, list(trace_dir.iterdir())
+

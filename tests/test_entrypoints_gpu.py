@@ -30,3 +30,61 @@ def test_torchrun_world1_snapshot_resume_on_gpu(tmp_path):
     r2 = _run(base + ["3", "1"], cwd=tmp_path)
     assert r2.returncode == 0, r2.stderr[-2000:]
     assert "Resuming training from snapshot at Epoch" in r2.stdout
+
+
+def _free_port():
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return str(s.getsockname()[1])
+
+
+def test_multigpu_spawn_world2_one_device(tmp_path):
+    """Pre-flight for the driver's 8-GPU run: the
+    mp.spawn entrypoint at world 2 with both ranks on ONE device. RCCL
+    refuses same-device ranks, so this also proves the entrypoint path's
+    all-ranks-agreed downgrade to gloo (create_comm -> build_gpu_comm)."""
+    r = _run([sys.executable, os.path.join(ROOT, "multigpu.py"), "2", "1"],
+             cwd=tmp_path,
+             env_extra={"MI355X_FORCE_DEV0": "1", "MI355X_WORLD": "2",
+                        "MASTER_PORT": _free_port()},
+             timeout=600)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert "Epoch 1" in r.stdout
+    assert (tmp_path / "checkpoint.pt").exists()
+
+
+def test_torchrun_world2_one_device_snapshot(tmp_path):
+    """Stage 3 (torchrun) at world 2 on one device — the exact launch form
+    the driver uses at N>1, rehearsed on a 1-GPU box."""
+    script = os.path.join(ROOT, "multigpu_torchrun.py")
+    base = TORCHRUN + ["--standalone", "--local-addr", "127.0.0.1",
+                       "--nproc_per_node", "2", script]
+    r = _run(base + ["2", "1"], cwd=tmp_path,
+             env_extra={"MI355X_FORCE_DEV0": "1"}, timeout=600)
+    assert r.returncode == 0, r.stderr[-3000:]
+    assert (tmp_path / "snapshot.pt").exists()
+
+
+def test_profile_stage_world2_one_device(tmp_path):
+    """Profile stage at world 2 on one device: per-rank TB traces AND the
+    per-rank kernel summaries come out of one command."""
+    r = _run([sys.executable, os.path.join(ROOT, "multigpu_profile.py"), "2"],
+             cwd=tmp_path,
+             env_extra={"MI355X_FORCE_DEV0": "1", "MI355X_WORLD": "2",
+                        "MASTER_PORT": _free_port(),
+                        "MI355X_PROFILE_DATASET": "128",
+                        "MI355X_PROFILE_BATCH": "16"},
+             timeout=900)
+    assert r.returncode == 0, r.stderr[-3000:]
+    trace_dir = tmp_path / "log" / "resnet50"
+    stats = sorted(trace_dir.glob("kernel_stats_rank*.json"))
+    assert len(stats) == 2, list(trace_dir.iterdir())
+    import json
+    rows = json.loads(stats[0].read_text())
+    assert rows and any("Conv" in r["name"] or "conv" in r["name"]
+                        or "igemm" in r["name"] or "miopen" in r["name"].lower()
+                        for r in rows[:15]), [r["name"] for r in rows[:15]]
+    traces = list(trace_dir.glob("*.pt.trace.json*")) \
+        + list(trace_dir.glob("*.json.gz"))
+    assert len([p for p in traces if "kernel_stats" not in p.name]) >= 2
