@@ -49,18 +49,28 @@ RcclComm::RcclComm(const std::string& unique_id, int rank, int world, int device
   ncclUniqueId id;
   memcpy(id.internal, unique_id.data(), NCCL_UNIQUE_ID_BYTES);
   HIP_OK(hipSetDevice(device));
-  // Dedicated, high-priority comm stream: bucket all-reduces launched here
-  // overlap with backward compute on the torch stream.
-  // MI355X_COMM_PRIO=0 drops the priority (diagnostic: high-priority
-  // streams were a suspect in the MIOpen find-time solution regression).
+  // Dedicated comm stream at priority -1 (torch's "high"): bucket
+  // all-reduces launched here overlap with backward compute on the torch
+  // stream. NOT the device's GREATEST priority: a live
+  // communicator whose idle comm stream sat at greatest priority during
+  // MIOpen's conv solution search made MIOpen pick ~2.5x slower conv
+  // kernels (44.3 vs 17.5 ms/step ResNet-50 — consistent with
+  // highest-priority HW queues reserving CUs and distorting find-time
+  // measurements; bisected on hardware, tools/miopen_comm_probe.py:
+  // greatest->44.3, default/-1->17.4-18.0). MI355X_COMM_PRIO: "0" =
+  // default priority, "greatest" = the old extreme (diagnostics).
   const char* prio_env = getenv("MI355X_COMM_PRIO");
   if (prio_env && strcmp(prio_env, "0") == 0) {
     HIP_OK(hipStreamCreateWithFlags(&comm_stream_, hipStreamNonBlocking));
   } else {
     int least = 0, greatest = 0;
     HIP_OK(hipDeviceGetStreamPriorityRange(&least, &greatest));
+    int prio = -1;
+    if (prio_env && strcmp(prio_env, "greatest") == 0) prio = greatest;
+    if (prio < greatest) prio = greatest;  // clamp into the valid range
+    if (prio > least) prio = least;
     HIP_OK(hipStreamCreateWithPriority(&comm_stream_, hipStreamNonBlocking,
-                                       greatest));
+                                       prio));
   }
   HIP_OK(hipEventCreateWithFlags(&ready_ev_, hipEventDisableTiming));
   HIP_OK(hipEventCreateWithFlags(&done_ev_, hipEventDisableTiming));

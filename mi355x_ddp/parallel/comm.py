@@ -75,25 +75,37 @@ def ddp_setup(rank: Optional[int] = None, world_size: Optional[int] = None,
 
 
 class GlooComm:
-    """CPU shadow communicator: same interface as the RCCL path, backed by
-    torch.distributed gloo. Async handles model the comm-stream overlap."""
+    """CPU shadow / fallback communicator: same interface as the RCCL
+    path, backed by torch.distributed gloo. Async handles model the
+    comm-stream overlap.
+
+    Collectives go through a DEDICATED gloo group, not the default group:
+    under the compound "cpu:gloo,cuda:nccl" backend (ddp_setup), a CUDA
+    tensor on the default group routes to ProcessGroupNCCL — which is
+    exactly the transport this fallback exists to avoid (e.g. same-device
+    ranks, where NCCL raises 'Duplicate GPU detected'). ProcessGroupGloo
+    handles CUDA tensors by host staging. All ranks construct GlooComm
+    together (the transport ladder agrees first), so the collective
+    new_group call is safe."""
 
     def __init__(self):
         assert dist.is_initialized()
         self.rank = dist.get_rank()
         self.world = dist.get_world_size()
+        self._pg = dist.new_group(backend="gloo")
         self._handles: List = []
 
     def all_reduce_avg(self, t: torch.Tensor) -> None:
-        h = dist.all_reduce(t, op=dist.ReduceOp.SUM, async_op=True)
+        h = dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self._pg,
+                            async_op=True)
         self._handles.append((h, t))
 
     def all_reduce_avg_inline(self, t: torch.Tensor) -> None:
-        dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self._pg)
         t.div_(self.world)
 
     def broadcast(self, t: torch.Tensor, root: int = 0) -> None:
-        dist.broadcast(t, src=root)
+        dist.broadcast(t, src=root, group=self._pg)
 
     def join_compute(self) -> None:
         for h, t in self._handles:
@@ -102,7 +114,7 @@ class GlooComm:
         self._handles.clear()
 
     def barrier(self) -> None:
-        dist.barrier()
+        dist.barrier(group=self._pg)
 
 
 class RcclCommAdapter:
