@@ -49,24 +49,25 @@ RcclComm::RcclComm(const std::string& unique_id, int rank, int world, int device
   ncclUniqueId id;
   memcpy(id.internal, unique_id.data(), NCCL_UNIQUE_ID_BYTES);
   HIP_OK(hipSetDevice(device));
-  // Dedicated comm stream at priority -1 (torch's "high"): bucket
-  // all-reduces launched here overlap with backward compute on the torch
-  // stream. NOT the device's GREATEST priority: a live
-  // communicator whose idle comm stream sat at greatest priority during
-  // MIOpen's conv solution search made MIOpen pick ~2.5x slower conv
-  // kernels (44.3 vs 17.5 ms/step ResNet-50 — consistent with
-  // highest-priority HW queues reserving CUs and distorting find-time
-  // measurements; bisected on hardware, tools/miopen_comm_probe.py:
-  // greatest->44.3, default/-1->17.4-18.0). MI355X_COMM_PRIO: "0" =
-  // default priority, "greatest" = the old extreme (diagnostics).
+  // Dedicated comm stream at DEFAULT priority: bucket all-reduces
+  // launched here overlap with backward compute on the torch stream
+  //  — measured 90%+ overlap at cap 25 MB without any
+  // priority elevation (tools/overlap_trace.py). Elevated priority is
+  // actively harmful here: a live communicator whose IDLE comm stream
+  // sits at raised priority (greatest OR -1) during MIOpen's conv
+  // solution search makes MIOpen pick ~2.5x slower conv kernels
+  // (44.3 vs 17.5-18.0 ms/step ResNet-50; bisected on hardware,
+  // tools/miopen_comm_probe.py + comm_order_debug.py — default priority
+  // is the only safe arm; stock ProcessGroupNCCL also defaults to
+  // non-elevated comm streams). MI355X_COMM_PRIO=<int|greatest> raises
+  // it for experiments.
   const char* prio_env = getenv("MI355X_COMM_PRIO");
-  if (prio_env && strcmp(prio_env, "0") == 0) {
+  if (prio_env == nullptr || strcmp(prio_env, "0") == 0) {
     HIP_OK(hipStreamCreateWithFlags(&comm_stream_, hipStreamNonBlocking));
   } else {
     int least = 0, greatest = 0;
     HIP_OK(hipDeviceGetStreamPriorityRange(&least, &greatest));
-    int prio = -1;
-    if (prio_env && strcmp(prio_env, "greatest") == 0) prio = greatest;
+    int prio = strcmp(prio_env, "greatest") == 0 ? greatest : atoi(prio_env);
     if (prio < greatest) prio = greatest;  // clamp into the valid range
     if (prio > least) prio = least;
     HIP_OK(hipStreamCreateWithPriority(&comm_stream_, hipStreamNonBlocking,
