@@ -143,6 +143,38 @@ def resnet_arms(args, results):
         oopt.step()
     results["ours-hooks"] = timeit(ours, args.steps, args.warmup)
 
+    # whole-step hipGraph capture of the ResNet step (eager fallback if
+    # MIOpen ops refuse capture — the arm then reports ~hooks numbers)
+    from mi355x_ddp.engine import GraphedAutogradStep
+
+    class _Autocast(torch.nn.Module):
+        def __init__(self, inner):
+            super().__init__()
+            self.inner = inner
+
+        def forward(self, xx):
+            with autocast():
+                return self.inner(xx)
+
+    torch.manual_seed(0)
+    gm = resnet50().to(dev).to(memory_format=fmt)
+    geng = DDP(gm, comm=None, cpp_hooks=False)
+    gopt = FusedSGD(gm.parameters(), lr=1e-4)
+    gopt.attach_reducer(geng.reducer)
+
+    def gloss(y, tt):
+        with autocast():
+            return ops.cross_entropy(y, tt)
+
+    gs = GraphedAutogradStep(_Autocast(geng), gloss, gopt,
+                             finalize=geng.finalize_backward,
+                             warmup_steps=2)
+    # trigger warmup+capture outside the timed region
+    gs.step(x, t)
+    results["ours-graph"] = timeit(lambda: gs.step(x, t),
+                                   args.steps, args.warmup)
+    results["graph_captured"] = bool(gs._graphs) and not gs._broken
+
 
 def main():
     ap = argparse.ArgumentParser()
