@@ -88,3 +88,38 @@ def test_profile_stage_world2_one_device(tmp_path):
     traces = list(trace_dir.glob("*.pt.trace.json*")) \
         + list(trace_dir.glob("*.json.gz"))
     assert len([p for p in traces if "kernel_stats" not in p.name]) >= 2
+
+
+def test_multinode_two_groups_of_four_one_device(tmp_path):
+    """BASELINE.json config 5 on GPU: 2 x 4-rank torchrun groups (c10d
+    rendezvous) forming world 8, all ranks time-sharing one device
+    (MI355X_FORCE_DEV0). Covers the compound cpu:gloo,cuda:nccl backend,
+    the agreed RCCL->gloo downgrade at world 8, and the multinode
+    entrypoint's MSE/snapshot path in the driver-like launch form."""
+    import subprocess
+    script = os.path.join(ROOT, "multinode_torchrun.py")
+    port = _free_port()
+    env = dict(os.environ)
+    env["PYTHONPATH"] = ROOT + os.pathsep + env.get("PYTHONPATH", "")
+    env.setdefault("OMP_NUM_THREADS", "1")
+    env["MI355X_FORCE_DEV0"] = "1"
+    procs = []
+    for node in range(2):
+        procs.append(subprocess.Popen(
+            TORCHRUN + ["--nnodes", "2", "--nproc_per_node", "4",
+                        "--node-rank", str(node), "--local-addr", "127.0.0.1",
+                        "--rdzv_backend", "c10d",
+                        "--rdzv_endpoint", f"127.0.0.1:{port}",
+                        "--rdzv_id", "gpujob8", script, "1", "1"],
+            cwd=tmp_path, env=env, stdout=subprocess.PIPE,
+            stderr=subprocess.STDOUT, text=True))
+    outs = []
+    for p in procs:
+        out, _ = p.communicate(timeout=540)
+        outs.append(out)
+    assert all(p.returncode == 0 for p in procs), outs[-1][-3000:]
+    allout = "\n".join(outs)
+    for r in range(8):
+        assert f"[GPU{r}] Epoch 0" in allout, allout[-2000:]
+    assert "Steps: 8" in allout  # 2048/8 ranks/32 batch (SURVEY §2.4)
+    assert (tmp_path / "snapshot.pt").exists()
