@@ -79,9 +79,16 @@ struct MseFn : public torch::autograd::Function<MseFn> {
 
   static variable_list backward(AutogradContext* ctx, variable_list grads) {
     auto saved = ctx->get_saved_variables();
-    auto dy = mse_bwd(saved[0], saved[1], 1.0);
-    // capture-safe: no D2H read of the incoming grad — scale on device
-    dy = dy * grads[0];
+    const auto& g0 = grads[0];
+    // capture-safe: no D2H read of the incoming grad — when it is the
+    // usual device f32 scalar (autograd's seed), fold it INTO the bwd
+    // kernel; otherwise scale with a separate elementwise multiply
+    if (g0.defined() && g0.is_cuda() && g0.numel() == 1 &&
+        g0.scalar_type() == at::kFloat) {
+      return {mse_bwd(saved[0], saved[1], 1.0, g0), torch::Tensor()};
+    }
+    auto dy = mse_bwd(saved[0], saved[1], 1.0, c10::nullopt);
+    dy = dy * g0;
     return {dy, torch::Tensor()};
   }
 };
@@ -96,8 +103,14 @@ struct CeFn : public torch::autograd::Function<CeFn> {
 
   static variable_list backward(AutogradContext* ctx, variable_list grads) {
     auto saved = ctx->get_saved_variables();
-    auto dy = ce_bwd(saved[0], saved[1], saved[2], 1.0);
-    dy = dy * grads[0];
+    const auto& g0 = grads[0];
+    if (g0.defined() && g0.is_cuda() && g0.numel() == 1 &&
+        g0.scalar_type() == at::kFloat) {  // fold the seed into the kernel
+      return {ce_bwd(saved[0], saved[1], saved[2], 1.0, g0),
+              torch::Tensor()};
+    }
+    auto dy = ce_bwd(saved[0], saved[1], saved[2], 1.0, c10::nullopt);
+    dy = dy * g0;
     return {dy, torch::Tensor()};
   }
 };

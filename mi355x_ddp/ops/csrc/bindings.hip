@@ -28,9 +28,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gemm_bf16", &mi355x::gemm_bf16, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt);
   m.def("ce_fwd", &mi355x::ce_fwd);
-  m.def("ce_bwd", &mi355x::ce_bwd);
+  m.def("ce_bwd", &mi355x::ce_bwd, py::arg("probs"), py::arg("t"),
+        py::arg("tsum"), py::arg("grad_scale"),
+        py::arg("gout") = c10::nullopt);
   m.def("mse_fwd", &mi355x::mse_fwd);
-  m.def("mse_bwd", &mi355x::mse_bwd);
+  m.def("mse_bwd", &mi355x::mse_bwd, py::arg("y"), py::arg("t"),
+        py::arg("grad_scale"), py::arg("gout") = c10::nullopt);
   m.def("sgd_flat", &mi355x::sgd_flat, py::arg("param_flat"),
         py::arg("grad_flat"), py::arg("lr"), py::arg("zero_grad") = true);
   m.def("build_copy_plan",

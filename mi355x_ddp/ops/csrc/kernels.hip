@@ -519,25 +519,38 @@ std::vector<torch::Tensor> ce_fwd(torch::Tensor y, torch::Tensor t) {
 template <typename T>
 __global__ void k_ce_bwd(const float* __restrict__ P, const T* __restrict__ Tg,
                          const float* __restrict__ tsum, T* __restrict__ dY,
-                         float scale, int B, int C) {
+                         float scale, int B, int C,
+                         const float* __restrict__ gout) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   const int64_t n = (int64_t)B * C;
   if (i >= n) return;
+  // gout: the incoming scalar grad (autograd's seed) read on-device —
+  // folds the dy*gout elementwise multiply into this kernel (one fewer
+  // launch per training step on the generic path)
+  const float s = gout ? scale * *gout : scale;
   const int row = (int)(i / C);
-  stf(&dY[i], (tsum[row] * P[i] - ldf(&Tg[i])) * scale);
+  stf(&dY[i], (tsum[row] * P[i] - ldf(&Tg[i])) * s);
 }
 
 torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor t,
-                     torch::Tensor tsum, double grad_scale) {
+                     torch::Tensor tsum, double grad_scale,
+                     c10::optional<torch::Tensor> gout) {
   const int B = (int)probs.size(0), C = (int)probs.size(1);
   auto tc = t.contiguous();
   auto dy = at::empty({B, C}, tc.options());
   const int64_t n = (int64_t)B * C;
+  const float* gp = nullptr;
+  if (gout.has_value()) {
+    TORCH_CHECK(gout->is_cuda() && gout->numel() == 1 &&
+                gout->scalar_type() == at::kFloat,
+                "ce_bwd gout must be a device f32 scalar");
+    gp = gout->data_ptr<float>();
+  }
   DISPATCH_F32_BF16(tc.scalar_type(), "ce_bwd", {
     hipLaunchKernelGGL((k_ce_bwd<scalar_t>), dim3(cdiv(n, 256)), dim3(256), 0,
                        cur_stream(), probs.data_ptr<float>(),
                        cdptr<scalar_t>(tc), tsum.data_ptr<float>(),
-                       dptr<scalar_t>(dy), (float)(grad_scale / B), B, C);
+                       dptr<scalar_t>(dy), (float)(grad_scale / B), B, C, gp);
   });
   HIP_OK(hipGetLastError());
   return dy;
@@ -592,20 +605,31 @@ torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t) {
 
 template <typename T>
 __global__ void k_mse_bwd(const T* __restrict__ Y, const T* __restrict__ Tg,
-                          T* __restrict__ dY, float scale, int64_t n) {
+                          T* __restrict__ dY, float scale, int64_t n,
+                          const float* __restrict__ gout) {
   const int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
-  if (i < n) stf(&dY[i], (ldf(&Y[i]) - ldf(&Tg[i])) * scale);
+  const float s = gout ? scale * *gout : scale;  // see k_ce_bwd
+  if (i < n) stf(&dY[i], (ldf(&Y[i]) - ldf(&Tg[i])) * s);
 }
 
-torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale) {
+torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale,
+                      c10::optional<torch::Tensor> gout) {
   auto yc = y.contiguous();
   auto tc = t.contiguous();
   const int64_t n = yc.numel();
   auto dy = at::empty_like(yc);
+  const float* gp = nullptr;
+  if (gout.has_value()) {
+    TORCH_CHECK(gout->is_cuda() && gout->numel() == 1 &&
+                gout->scalar_type() == at::kFloat,
+                "mse_bwd gout must be a device f32 scalar");
+    gp = gout->data_ptr<float>();
+  }
   DISPATCH_F32_BF16(y.scalar_type(), "mse_bwd", {
     hipLaunchKernelGGL((k_mse_bwd<scalar_t>), dim3(cdiv(n, 256)), dim3(256), 0,
                        cur_stream(), cdptr<scalar_t>(yc), cdptr<scalar_t>(tc),
-                       dptr<scalar_t>(dy), (float)(2.0 * grad_scale / n), n);
+                       dptr<scalar_t>(dy), (float)(2.0 * grad_scale / n), n,
+                       gp);
   });
   HIP_OK(hipGetLastError());
   return dy;

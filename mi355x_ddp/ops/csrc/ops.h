@@ -30,12 +30,16 @@ torch::Tensor gemm_bf16(torch::Tensor x, torch::Tensor w,
 // degenerate Linear(20,1) case, single_gpu.py:24 + utils.py:7).
 // Returns (loss[scalar], probs[B,C], tsum[B]).
 std::vector<torch::Tensor> ce_fwd(torch::Tensor y, torch::Tensor t);
+// gout: optional device f32 scalar (autograd's incoming grad) folded
+// into the kernel — saves the separate dy*gout launch per step.
 torch::Tensor ce_bwd(torch::Tensor probs, torch::Tensor t,
-                     torch::Tensor tsum, double grad_scale);
+                     torch::Tensor tsum, double grad_scale,
+                     c10::optional<torch::Tensor> gout = c10::nullopt);
 
 // MSE (reference multinode_torchrun.py:46). Returns loss[scalar].
 torch::Tensor mse_fwd(torch::Tensor y, torch::Tensor t);
-torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale);
+torch::Tensor mse_bwd(torch::Tensor y, torch::Tensor t, double grad_scale,
+                      c10::optional<torch::Tensor> gout = c10::nullopt);
 
 // Fused SGD over a flat bucket: p -= lr*g; optionally g = 0 in the same
 // kernel (folds reference optimizer.step() + zero_grad, single_gpu.py:22,26).
