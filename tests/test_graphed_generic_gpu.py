@@ -210,3 +210,46 @@ def test_graphed_shard_bound_matches_eager_bitwise():
     torch.cuda.synchronize()
     for a, b in zip(m_g.parameters(), m_e.parameters()):
         assert torch.equal(a, b), (a - b).abs().max()
+
+
+def test_graphed_toy_step_with_world1_rccl():
+    # GraphedToyStep capture under a communicator was
+    # untested. At world 1 the RCCL inline all-reduce is the identity, so
+    # the captured toy step must match the comm-less fused step bitwise.
+    import torch.distributed as dist
+
+    from mi355x_ddp.engine import GraphedToyStep, ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29785")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        from mi355x_ddp.parallel.comm import RcclCommAdapter
+        comm = RcclCommAdapter(torch.device(DEV))
+        g = torch.Generator().manual_seed(41)
+        X = torch.rand(8, 32, 20, generator=g).to(DEV)
+        T = torch.rand(8, 32, 1, generator=g).to(DEV)
+
+        torch.manual_seed(9)
+        m_g = toy_model(20, 1).to(DEV)
+        eng_g = GraphedToyStep(m_g, comm=comm, lr=0.03, use_mse=True)
+        for s in range(8):
+            eng_g.step(X[s].contiguous(), T[s].contiguous())
+        torch.cuda.synchronize()
+        assert eng_g._graph not in (None, False), \
+            "capture with the inline RCCL collective must succeed"
+
+        torch.manual_seed(9)
+        m_e = toy_model(20, 1).to(DEV)
+        eng_e = ToyFusedStep(m_e, comm=None, lr=0.03, use_mse=True)
+        # replicate the capture call's accounting: call 0 runs one eager
+        # step (warmup) and the captured step is NOT executed that call
+        eng_e.step(X[0].contiguous(), T[0].contiguous())
+        for s in range(1, 8):
+            eng_e.step(X[s].contiguous(), T[s].contiguous())
+        torch.cuda.synchronize()
+        assert torch.allclose(m_g.weight.detach(), m_e.weight.detach(),
+                              atol=1e-6), \
+            (m_g.weight - m_e.weight).abs().max()
+    finally:
+        dist.destroy_process_group()
