@@ -1758,8 +1758,9 @@ void toy_multistep_mesh(torch::Tensor x, torch::Tensor t,
                         bool use_mse, int64_t w_off, int64_t b_off, double lr,
                         int64_t batch, P2pMesh& mesh) {
   const int B = (int)batch, K = (int)x.size(1);
-  TORCH_CHECK(B == 32 && K == 20,
-              "mesh multistep supports the reference shape (batch 32, K 20)");
+  TORCH_CHECK((B == 32 || B == 64) && K == 20,
+              "mesh multistep supports the fast-path shapes "
+              "(batch 32 or 64, K 20)");
   TORCH_CHECK(x.is_contiguous() && t.is_contiguous());
   TORCH_CHECK(x.size(0) % B == 0 && t.size(0) == x.size(0));
   TORCH_CHECK(lr > 0.0, "mesh multistep applies SGD in-kernel");
@@ -1773,20 +1774,27 @@ void toy_multistep_mesh(torch::Tensor x, torch::Tensor t,
   const unsigned long long seq0 = mesh.alloc_seq((unsigned long long)S);
   const float inv = 1.f / (float)mesh.world();
   if (x.scalar_type() == at::kFloat) {
-    hipLaunchKernelGGL((k_toy_multistep_spec<float, 32, 20, true>), dim3(1),
-                       dim3(64), 0, cur_stream(), cdptr<float>(x),
-                       cdptr<float>(t), dptr<float>(param_flat), lossp, S,
-                       use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
-                       mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
-                       seq0, mesh.err_flag());
+    auto go = [&](auto kptr) {
+      hipLaunchKernelGGL(kptr, dim3(1), dim3(64), 0, cur_stream(),
+                         cdptr<float>(x), cdptr<float>(t),
+                         dptr<float>(param_flat), lossp, S,
+                         use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
+                         mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
+                         seq0, mesh.err_flag());
+    };
+    if (B == 32) go(k_toy_multistep_spec<float, 32, 20, true>);
+    else go(k_toy_multistep_spec<float, 64, 20, true>);
   } else if (x.scalar_type() == at::kBFloat16) {
-    hipLaunchKernelGGL((k_toy_multistep_bf16w<32, 20, true>), dim3(1),
-                       dim3(64), 0, cur_stream(),
-                       cdptr<__hip_bfloat16>(x), cdptr<__hip_bfloat16>(t),
-                       dptr<__hip_bfloat16>(param_flat), lossp, S,
-                       use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
-                       mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
-                       seq0, mesh.err_flag());
+    auto go = [&](auto kptr) {
+      hipLaunchKernelGGL(kptr, dim3(1), dim3(64), 0, cur_stream(),
+                         cdptr<__hip_bfloat16>(x), cdptr<__hip_bfloat16>(t),
+                         dptr<__hip_bfloat16>(param_flat), lossp, S,
+                         use_mse ? 1 : 0, (int)w_off, (int)b_off, (float)lr,
+                         mesh.peer_slots(), mesh.my_mb(), mesh.world(), inv,
+                         seq0, mesh.err_flag());
+    };
+    if (B == 32) go(k_toy_multistep_bf16w<32, 20, true>);
+    else go(k_toy_multistep_bf16w<64, 20, true>);
   } else {
     TORCH_CHECK(false, "mesh multistep: dtype must be f32 or bf16");
   }

@@ -257,3 +257,79 @@ def test_trainer_persistent_mesh_world2(tmp_path):
              nprocs=2, join=True)
     w = torch.load(tmp_path / "trainer_mesh.pt", weights_only=True)["w"]
     assert torch.isfinite(w).all() and w.abs().sum() > 0
+
+
+def _persistent_b64_worker(rank, world, port, dtype_name, out_dir):
+    # batch-64 instantiations of the MESH multistep kernels (r02: the
+    # fast-path matrix covers B in {32,64} x {f32,bf16} at any world)
+    from mi355x_ddp.engine import PersistentToyStep
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel.comm import GlooComm, P2pMeshComm
+    _init(rank, world, port)
+    dtype = torch.bfloat16 if dtype_name == "bf16" else torch.float32
+    try:
+        comm = P2pMeshComm(torch.device("cuda", 0), base=GlooComm())
+        torch.manual_seed(13)
+        model = toy_model(20, 1).to("cuda").to(dtype)
+        eng = PersistentToyStep(model, comm=comm, lr=0.04, use_mse=True)
+        eng.reducer.broadcast_params(root=0)
+        g = torch.Generator().manual_seed(200 + rank)
+        Xf = torch.rand(6 * 64, 20, generator=g).to("cuda").to(dtype)
+        Tf = torch.rand(6 * 64, 1, generator=g).to("cuda").to(dtype)
+        eng.bind_shard(Xf, Tf, 64)
+        for s in range(6):
+            eng.step_shard(s)
+        eng.flush()
+        torch.cuda.synchronize()
+        comm.check()
+        if rank == 0:
+            torch.save({"w": model.weight.detach().float().cpu(),
+                        "b": model.bias.detach().float().cpu()},
+                       os.path.join(out_dir, f"pm64_{dtype_name}.pt"))
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+def _fused_b64_worker(rank, world, port, dtype_name, out_dir):
+    from mi355x_ddp.engine import ToyFusedStep
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel.comm import GlooComm
+    _init(rank, world, port)
+    dtype = torch.bfloat16 if dtype_name == "bf16" else torch.float32
+    try:
+        comm = GlooComm()
+        torch.manual_seed(13)
+        model = toy_model(20, 1).to("cuda").to(dtype)
+        eng = ToyFusedStep(model, comm=comm, lr=0.04, use_mse=True)
+        eng.reducer.broadcast_params(root=0)
+        g = torch.Generator().manual_seed(200 + rank)
+        Xf = torch.rand(6 * 64, 20, generator=g).to("cuda").to(dtype)
+        Tf = torch.rand(6 * 64, 1, generator=g).to("cuda").to(dtype)
+        for s in range(6):
+            eng.step(Xf[s * 64:(s + 1) * 64].contiguous(),
+                     Tf[s * 64:(s + 1) * 64].contiguous())
+        torch.cuda.synchronize()
+        if rank == 0:
+            torch.save({"w": model.weight.detach().float().cpu(),
+                        "b": model.bias.detach().float().cpu()},
+                       os.path.join(out_dir, f"fused64_{dtype_name}.pt"))
+        torch.distributed.barrier()
+    finally:
+        torch.distributed.destroy_process_group()
+
+
+@pytest.mark.parametrize("dtype_name", ["fp32", "bf16"])
+def test_persistent_mesh_world2_batch64(dtype_name, tmp_path):
+    mp.spawn(_fused_b64_worker,
+             args=(2, _free_port(), dtype_name, str(tmp_path)),
+             nprocs=2, join=True)
+    mp.spawn(_persistent_b64_worker,
+             args=(2, _free_port(), dtype_name, str(tmp_path)),
+             nprocs=2, join=True)
+    a = torch.load(tmp_path / f"fused64_{dtype_name}.pt", weights_only=True)
+    b = torch.load(tmp_path / f"pm64_{dtype_name}.pt", weights_only=True)
+    tol = 5e-3 if dtype_name == "bf16" else 1e-5
+    assert torch.allclose(a["w"], b["w"], atol=tol, rtol=5e-2), \
+        (a["w"] - b["w"]).abs().max()
+    assert torch.allclose(a["b"], b["b"], atol=tol, rtol=5e-2)

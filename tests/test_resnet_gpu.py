@@ -108,3 +108,36 @@ def test_vit_l32_train_step_multibucket():
         opt.step()
         l0 = l0 or float(loss.detach())
     assert torch.isfinite(loss.detach()) and float(loss.detach()) != l0
+
+
+def test_trainer_autocast_bf16_nhwc_knobs(monkeypatch, tmp_path):
+    # the measured-fastest ResNet configuration (profiles r02d: bf16
+    # autocast + channels_last) through the Trainer env knobs
+    import os
+    import torch.distributed as dist
+
+    from mi355x_ddp.data import RandomImageDataset, prepare_dataloader
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.trainer import Trainer
+    monkeypatch.setenv("MI355X_AUTOCAST_BF16", "1")
+    monkeypatch.setenv("MI355X_NHWC", "1")
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29786")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        torch.manual_seed(0)
+        model = resnet50()
+        data = prepare_dataloader(RandomImageDataset(32, (3, 224, 224)), 16)
+        opt = FusedSGD(model.parameters(), lr=1e-4)
+        tr = Trainer(model, data, opt, gpu_id=0, save_every=10**9,
+                     checkpoint_path=str(tmp_path / "ck.pt"), loss_fn="ce")
+        assert tr._autocast and tr._nhwc
+        tr.train(1)
+        torch.cuda.synchronize()
+        m = tr._unwrapped()
+        assert all(torch.isfinite(p).all() for p in m.parameters())
+        # params stayed fp32 (autocast, not a model cast); NHWC applied
+        assert next(m.parameters()).dtype == torch.float32
+        assert m.conv1.weight.is_contiguous(memory_format=torch.channels_last)
+    finally:
+        dist.destroy_process_group()
