@@ -57,7 +57,10 @@ def _library_gemm_shape(k: int, n: int) -> bool:
     """Plain library-GEMM territory (north star: hipBLASLt/rocBLAS only
     for plain library GEMMs): measured crossover on MI355X —
     32x2048@2048x1000 runs ~1.8x faster through rocBLAS, while our MFMA
-    kernels win up to ~256x256 contractions (profiles/kernel_bench)."""
+    kernels win up to ~256x256 contractions (profiles/kernel_bench).
+    The dispatch itself lives in C++ now (csrc/autograd_ops.hip
+    library_gemm_shape — keep the two in sync); this mirror is the
+    documented constant and is used by tests."""
     return k * n >= (1 << 20)
 
 
