@@ -33,7 +33,19 @@ class DDP(torch.nn.Module):
                  bucket_cap_mb: Optional[float] = None,
                  comm=None, grad_views: Optional[bool] = None,
                  broadcast_buffers: bool = True,
-                 cpp_hooks: Optional[bool] = None):
+                 cpp_hooks: Optional[bool] = None,
+                 find_unused_parameters: bool = False,
+                 gradient_as_bucket_view: Optional[bool] = None,
+                 static_graph: bool = False):
+        # Stock-DDP drop-in kwargs: find_unused_parameters is a no-op —
+        # models with conditionally-unused params train here BY DEFAULT
+        # (finalize() launches straggler buckets; stock hangs without the
+        # flag). gradient_as_bucket_view maps onto grad_views (views are
+        # already our default). static_graph is accepted and ignored (an
+        # optimization hint for stock's reducer rebuild).
+        if gradient_as_bucket_view is not None and grad_views is None:
+            grad_views = gradient_as_bucket_view
+        del find_unused_parameters, static_graph
         # device_ids/output_device: accepted for drop-in compatibility with
         # the stock signature the reference uses (`DDP(model,
         # device_ids=[gpu_id])`, ref multigpu.py:36). The model must already

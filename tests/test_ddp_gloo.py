@@ -402,3 +402,16 @@ def test_debug_sync_catches_step_before_finalize(tmp_path):
              nprocs=WORLD, join=True)
     for r in range(WORLD):
         assert (tmp_path / f"race{r}.ok").read_text() == "caught"
+
+
+def test_ddp_accepts_stock_kwargs():
+    # full drop-in surface: stock DDP's remaining kwargs are accepted
+    # (find_unused_parameters/static_graph no-ops — unused params train
+    # by default here; gradient_as_bucket_view maps to grad_views)
+    m = _make_model(1)
+    eng = DDP(m, find_unused_parameters=True, static_graph=True,
+              gradient_as_bucket_view=True, broadcast_buffers=True)
+    assert eng.reducer.grad_views is True
+    m2 = _make_model(1)
+    eng2 = DDP(m2, gradient_as_bucket_view=False)
+    assert eng2.reducer.grad_views is False
