@@ -204,3 +204,30 @@ def test_ce_loss_class_index_targets_match_torch():
     y2 = y.detach().clone().requires_grad_(True)
     torch.nn.functional.cross_entropy(y2, idx).backward()
     assert torch.allclose(y.grad, y2.grad, atol=1e-6)
+
+
+def test_hooks_graph_engine_falls_back_on_cpu(tmp_path):
+    # engine="hooks-graph" is a GPU optimization; on CPU the Trainer runs
+    # the plain generic loop (no engine object) and trains identically
+    import torch.distributed as dist
+
+    from mi355x_ddp.data import MyTrainDataset, prepare_dataloader
+    from mi355x_ddp.models import toy_model
+    from mi355x_ddp.parallel import FusedSGD
+    from mi355x_ddp.trainer import Trainer
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29787")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        torch.manual_seed(3)
+        model = toy_model(20, 1)
+        data = prepare_dataloader(MyTrainDataset(64), 32)
+        opt = FusedSGD(model.parameters(), lr=1e-3)
+        tr = Trainer(model, data, opt, gpu_id="cpu", save_every=1,
+                     checkpoint_path=str(tmp_path / "c.pt"),
+                     loss_fn="mse", engine="hooks-graph")
+        assert tr._engine is None  # no graph engine off-GPU
+        tr.train(2)
+        assert (tmp_path / "c.pt").exists()
+    finally:
+        dist.destroy_process_group()
