@@ -11,6 +11,7 @@ buckets when the model is engine-wrapped.
 
 from __future__ import annotations
 
+import os
 from typing import Iterable
 
 import torch
@@ -49,7 +50,6 @@ class FusedSGD(torch.optim.Optimizer):
         # collectives were launched but finalize() never fenced them, that
         # read races the comm stream. Catch the protocol violation loudly
         # instead of training on half-reduced gradients.
-        import os
         if (os.environ.get("MI355X_DEBUG_SYNC") == "1"
                 and getattr(self, "_reducer", None) is not None
                 and self._reducer.unfenced):
