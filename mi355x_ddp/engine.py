@@ -242,8 +242,11 @@ class GraphedAutogradStep:
 
     Requirements: static shapes (same batch size every step) and a model
     without data-dependent control flow — the reference workloads (toy
-    Linear, ResNet-50) qualify. Capture failure falls back to eager with a
-    warning; results are unchanged either way.
+    Linear, ResNet-50) qualify. Autocast regions inside the step must use
+    cache_enabled=False (torch's AMP+graphs rule — the cast cache frees
+    tensors between iterations and invalidates the capture). Capture
+    failure falls back to eager with a warning; results are unchanged
+    either way.
     """
 
     def __init__(self, model, loss_fn, optimizer, finalize=None,

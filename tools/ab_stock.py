@@ -114,8 +114,12 @@ def resnet_arms(args, results):
         .contiguous(memory_format=fmt)
     t = torch.rand(args.batch, 1000, device=dev)
 
-    def autocast():
-        return torch.autocast("cuda", dtype=torch.bfloat16, enabled=bf16)
+    def autocast(cache=True):
+        # cache_enabled=False is REQUIRED under graph capture (torch's
+        # AMP+graphs rule: the autocast cast-cache frees its tensors
+        # between iterations, which invalidates a capture)
+        return torch.autocast("cuda", dtype=torch.bfloat16, enabled=bf16,
+                              cache_enabled=cache)
 
     torch.manual_seed(0)
     m = resnet50().to(dev).to(memory_format=fmt)
@@ -153,7 +157,7 @@ def resnet_arms(args, results):
             self.inner = inner
 
         def forward(self, xx):
-            with autocast():
+            with autocast(cache=False):
                 return self.inner(xx)
 
     torch.manual_seed(0)
@@ -163,7 +167,7 @@ def resnet_arms(args, results):
     gopt.attach_reducer(geng.reducer)
 
     def gloss(y, tt):
-        with autocast():
+        with autocast(cache=False):
             return ops.cross_entropy(y, tt)
 
     gs = GraphedAutogradStep(_Autocast(geng), gloss, gopt,
