@@ -324,6 +324,12 @@ class GraphedAutogradStep:
                 "GraphedAutogradStep running eager (correct, slower)",
                 RuntimeWarning, stacklevel=2)
             self._broken = True
+            # a failed capture latches hipErrorStreamCaptureInvalidated in
+            # HIP's per-thread error state; pop it or the NEXT eager launch
+            # re-reports it (observed with MIOpen NHWC workspace allocs
+            # invalidating the capture)
+            if ops.has_ext():
+                ops.ext().clear_hip_errors()
             torch.cuda.synchronize()
             return None
 

@@ -19,6 +19,10 @@ namespace py = pybind11;
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native kernels and RCCL communicator for mi355x_ddp";
 
+  // pop HIP's latched per-thread error (e.g. hipErrorStreamCaptureInvalidated
+  // after a failed graph capture) so eager fallback paths can proceed
+  m.def("clear_hip_errors", [] { (void)hipGetLastError(); });
+
   m.def("linear_fwd", &mi355x::linear_fwd, py::arg("x"), py::arg("w"),
         py::arg("bias") = c10::nullopt);
   m.def("linear_bwd_weight", &mi355x::linear_bwd_weight, py::arg("x"),
