@@ -30,6 +30,30 @@ from .models.toy import HipLinear
 from .parallel.reducer import Reducer
 
 
+def _recover_failed_capture(ctx):
+    """Best-effort cleanup after an INVALIDATED stream capture (e.g. a
+    library allocating outside torch's allocator mid-capture — observed
+    with MIOpen NHWC workspace allocs). torch.cuda.graph.__exit__ raises
+    from capture_end BEFORE exiting its stream context or ending the
+    allocator's route-to-pool state, leaving the thread on the capture
+    stream, allocations routed to the dead graph pool, and
+    hipErrorStreamCaptureInvalidated latched in HIP's thread state.
+    Undo all three so the eager fallback can proceed."""
+    try:
+        ctx.stream_ctx.__exit__(None, None, None)
+    except Exception:
+        pass
+    try:
+        dev = torch.cuda.current_device()
+        torch._C._cuda_endAllocateToPool(dev, ctx.pool[0])
+        torch._C._cuda_releasePool(dev, ctx.pool[0])
+    except Exception:
+        pass
+    if ops.has_ext():
+        ops.ext().clear_hip_errors()  # pop the latched HIP error
+    torch.cuda.synchronize()
+
+
 class ToyFusedStep:
     """One-kernel fwd+bwd for HipLinear(K,1) + MSE/CE, bucket all-reduce,
     fused SGD. Works for any world size (comm=None -> single process)."""
@@ -309,9 +333,10 @@ class GraphedAutogradStep:
                            dtype=self._bdtype[0], device=self._bdev)
         tbuf = torch.empty((G * B,) + self._bshape[1][1:],
                            dtype=self._bdtype[1], device=self._bdev)
+        g = torch.cuda.CUDAGraph()
+        ctx = torch.cuda.graph(g)
         try:
-            g = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(g):
+            with ctx:
                 for i in range(G):
                     self.loss = self._eager_step(xbuf[i * B:(i + 1) * B],
                                                  tbuf[i * B:(i + 1) * B])
@@ -324,13 +349,7 @@ class GraphedAutogradStep:
                 "GraphedAutogradStep running eager (correct, slower)",
                 RuntimeWarning, stacklevel=2)
             self._broken = True
-            # a failed capture latches hipErrorStreamCaptureInvalidated in
-            # HIP's per-thread error state; pop it or the NEXT eager launch
-            # re-reports it (observed with MIOpen NHWC workspace allocs
-            # invalidating the capture)
-            if ops.has_ext():
-                ops.ext().clear_hip_errors()
-            torch.cuda.synchronize()
+            _recover_failed_capture(ctx)
             return None
 
     def _note_shapes(self, x, t):
@@ -424,10 +443,12 @@ class GraphedToyStep(ToyFusedStep):
         # warm up collectives/kernels outside capture first
         super().step(self._x_static, self._t_static)
         torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        ctx = torch.cuda.graph(g)
         try:
-            self._graph = torch.cuda.CUDAGraph()
-            with torch.cuda.graph(self._graph):
+            with ctx:
                 super().step(self._x_static, self._t_static)
+            self._graph = g
         except Exception as e:  # capture unsupported -> eager fallback
             import warnings
             warnings.warn(
@@ -435,7 +456,7 @@ class GraphedToyStep(ToyFusedStep):
                 "GraphedToyStep running eager (correct, slower)",
                 RuntimeWarning, stacklevel=2)
             self._graph = False
-            torch.cuda.synchronize()
+            _recover_failed_capture(ctx)
 
     def step(self, x: torch.Tensor, t: torch.Tensor) -> None:
         if self._graph is None:
