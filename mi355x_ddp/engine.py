@@ -382,6 +382,14 @@ class GraphedAutogradStep:
         assert xs.is_cuda and xs.is_contiguous() and ts.is_contiguous()
         self._shard_x, self._shard_t, self._sbatch = xs, ts, batch
         self._s_start = self._s_next = 0
+        if (self._bshape is not None
+                and (xs[:batch].shape, ts[:batch].shape) != self._bshape):
+            # batch shape changed across binds: the captured graphs read
+            # fixed-shape static buffers — drop them and recapture lazily
+            # at the new shape (correct, pays one recapture)
+            torch.cuda.synchronize()
+            self._graphs.clear()
+            self._bshape = None
         self._note_shapes(xs[:batch], ts[:batch])
 
     def step_shard(self, i):

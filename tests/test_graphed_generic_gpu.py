@@ -253,3 +253,46 @@ def test_graphed_toy_step_with_world1_rccl():
             (m_g.weight - m_e.weight).abs().max()
     finally:
         dist.destroy_process_group()
+
+
+def test_graphed_shard_rebind_with_new_batch_size():
+    # binding a shard with a DIFFERENT batch size drops the captured
+    # graphs and recaptures at the new shape (stale fixed-shape buffers
+    # would otherwise corrupt the copy); training stays correct
+    m, eng, opt = _build(7)
+    gs = GraphedAutogradStep(eng, ops.mse_loss, opt,
+                             finalize=eng.finalize_backward,
+                             chunk_sizes=(4, 1))
+    g1 = torch.Generator().manual_seed(71)
+    xs1 = torch.randn(4 * 16, 64, generator=g1).to(DEV)
+    ts1 = torch.randn(4 * 16, 10, generator=g1).to(DEV)
+    gs.bind_shard(xs1, ts1, 16)
+    for i in range(4):
+        gs.step_shard(i)
+    gs.flush()
+    xs2 = torch.randn(4 * 8, 64, generator=g1).to(DEV)
+    ts2 = torch.randn(4 * 8, 10, generator=g1).to(DEV)
+    gs.bind_shard(xs2, ts2, 8)  # batch 16 -> 8: graphs invalidated
+    for i in range(4):
+        gs.step_shard(i)
+    gs.flush()
+    torch.cuda.synchronize()
+    assert not gs._broken and 4 in gs._graphs
+
+    m_e, eng_e, opt_e = _build(7)
+
+    def eager(x, t):
+        loss = ops.mse_loss(eng_e(x), t)
+        loss.backward()
+        eng_e.finalize_backward()
+        opt_e.step()
+
+    for _ in range(gs.warmup_steps):
+        eager(xs1[:16], ts1[:16])
+    for i in range(4):
+        eager(xs1[i * 16:(i + 1) * 16], ts1[i * 16:(i + 1) * 16])
+    for i in range(4):
+        eager(xs2[i * 8:(i + 1) * 8], ts2[i * 8:(i + 1) * 8])
+    torch.cuda.synchronize()
+    for a, b in zip(m.parameters(), m_e.parameters()):
+        assert torch.equal(a, b), (a - b).abs().max()
