@@ -84,7 +84,8 @@ class Trainer:
         reference semantics for arbitrary models), "hooks-graph" (the same
         generic path captured in one hipGraph and replayed — fast for any
         static-shape model; falls back to eager hooks if capture fails or
-        off-GPU), or "fused"/"persistent"/"graph"/"auto" to run the toy
+        off-GPU; NOTE: capture warmup performs a few extra real training
+        steps on the first batch, per torch's capture recipe), or "fused"/"persistent"/"graph"/"auto" to run the toy
         fast path (single-launch fused step / multi-step deferred kernel /
         hipGraph replay) when the model+loss+optimizer qualify; silently
         falls back to hooks otherwise. Unknown values raise."""

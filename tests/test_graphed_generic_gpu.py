@@ -296,3 +296,50 @@ def test_graphed_shard_rebind_with_new_batch_size():
     torch.cuda.synchronize()
     for a, b in zip(m.parameters(), m_e.parameters()):
         assert torch.equal(a, b), (a - b).abs().max()
+
+
+def test_graphed_recovers_from_invalidated_capture():
+    # force a capture invalidation deterministically (a synchronize mid-
+    # capture is illegal) and verify the engine recovers: warning, eager
+    # fallback, training continues and stays correct (engine.py
+    # _recover_failed_capture — exercised for real by MIOpen NHWC
+    # workspace allocs, profiles r02d)
+    import warnings as warnings_mod
+
+    m, eng, opt = _build(8)
+    state = {"broke": False}
+
+    def poisoned_loss(y, t):
+        if torch.cuda.is_current_stream_capturing() and not state["broke"]:
+            state["broke"] = True
+            torch.cuda.synchronize()  # invalidates the capture
+        return ops.mse_loss(y, t)
+
+    gs = GraphedAutogradStep(eng, poisoned_loss, opt,
+                             finalize=eng.finalize_backward)
+    data = _batches(4, seed=91)
+    with warnings_mod.catch_warnings(record=True) as w:
+        warnings_mod.simplefilter("always")
+        for x, t in data:
+            gs.step(x, t)
+    torch.cuda.synchronize()
+    assert gs._broken
+    assert any("capture failed" in str(x.message) for x in w)
+
+    # eager arm: identical schedule (warmup trains, the poisoned capture
+    # executed nothing, every step() ran eager)
+    m_e, eng_e, opt_e = _build(8)
+
+    def eager(x, t):
+        loss = ops.mse_loss(eng_e(x), t)
+        loss.backward()
+        eng_e.finalize_backward()
+        opt_e.step()
+
+    for _ in range(gs.warmup_steps):
+        eager(*data[0])
+    for x, t in data:
+        eager(x, t)
+    torch.cuda.synchronize()
+    for a, b in zip(m.parameters(), m_e.parameters()):
+        assert torch.equal(a, b), (a - b).abs().max()
