@@ -21,15 +21,39 @@ MI355X_EPOCH_BLOCK      epochs gathered per epoch_shard_multi launch in
                         boundaries (default: fills the engine's max_defer
                         window; 1 = per-epoch gathers)
 MI355X_ENGINE           fast-engine selection for the entrypoint scripts
-                        (hooks default | auto | fused | persistent |
-                        graph; silently falls back to hooks when the
-                        stage does not qualify)
+                        (hooks default | hooks-graph | auto | fused |
+                        persistent | graph; silently falls back to hooks
+                        when the stage does not qualify)
 MI355X_DTYPE            bf16 = cast the model in multigpu.py (BASELINE
                         config 2); unset = the reference's fp32
+MI355X_CPP_HOOKS        0 = Python reducer hooks instead of the C++
+                        ReducerCore on the GPU views path (default 1;
+                        graph capture forces the Python path internally —
+                        C++ node post-hooks segfault capture_end)
+MI355X_DEBUG_SYNC       1 = stream-race assertions:
+                        FusedSGD refuses unfenced bucket reads; the mesh
+                        verifies size routing across ranks per call
+MI355X_COMM_PRIO        comm-stream priority: unset/0 = DEFAULT priority
+                        (required — any elevated priority on a live
+                        communicator corrupts MIOpen's conv solution
+                        search, profiles r02c); <int>|greatest for
+                        experiments only
+MI355X_AUTOCAST_BF16    1 = Trainer runs forward+loss under bf16 autocast
+MI355X_NHWC             1 = Trainer converts model+batches to
+                        channels_last (with AUTOCAST_BF16: the fastest
+                        measured ResNet-50 config, profiles r02d)
+MI355X_FORCE_DEV0       test-only: run a multi-rank world on ONE device
+                        (IPC time-sharing) — the 8-GPU pre-flight
+                        rehearsals; never set in production
+MI355X_WORLD            world size for the spawn entrypoints off-GPU and
+                        under MI355X_FORCE_DEV0
+MI355X_CORE_DEBUG       reducer-core bisection aid (noop|norebind) — not
+                        a production knob
 
 RCCL's own tuning envs (NCCL_ALGO, NCCL_PROTO, NCCL_MIN/MAX_NCHANNELS)
 pass straight through to the large-bucket collective path — the knobs
-SURVEY §5.8 names for per-size algorithm selection over xGMI.
+SURVEY §5.8 names for per-size algorithm selection over xGMI
+(tools/bucket_sweep.py drives the sweep).
 """
 
 import os
