@@ -217,3 +217,46 @@ def test_core_debug_sync_catches_unfenced_step(monkeypatch):
         torch.cuda.synchronize()
     finally:
         dist.destroy_process_group()
+
+
+def test_core_straggler_and_no_sync_with_rccl_comm():
+    # finalize's straggler branch and skip_comm both exercised with a REAL
+    # communicator (world-1 RCCL: averages are identities, so parity vs
+    # the comm-less run holds)
+    import torch.distributed as dist
+    os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
+    os.environ.setdefault("MASTER_PORT", "29788")
+    dist.init_process_group("gloo", rank=0, world_size=1)
+    try:
+        from mi355x_ddp.parallel.comm import RcclCommAdapter
+        comm = RcclCommAdapter(torch.device(DEV))
+
+        def run(with_comm):
+            model = _Unused().to(DEV)
+            red = Reducer(list(model.parameters()),
+                          comm=comm if with_comm else None,
+                          bucket_cap_mb=0.001, grad_views=True)
+            assert red._core is not None
+            opt = FusedSGD(model.parameters(), lr=0.01)
+            opt.attach_reducer(red)
+            g = torch.Generator().manual_seed(31)
+            x = torch.randn(16, 64, generator=g).to(DEV)
+            t = torch.randn(16, 10, generator=g).to(DEV)
+            for step in range(4):
+                if step == 1:
+                    red.skip_comm = True  # accumulation step
+                torch.nn.functional.mse_loss(model(x), t).backward()
+                red.finalize()
+                if step == 1:
+                    red.skip_comm = False
+                    continue  # no optimizer step mid-accumulation
+                opt.step()
+            torch.cuda.synchronize()
+            return [p.detach().clone() for p in model.parameters()]
+
+        got = run(True)     # stragglers: dead layer's buckets all-reduce
+        want = run(False)   # in finalize; skip_comm honored at step 1
+        for a, b in zip(got, want):
+            assert torch.equal(a, b), (a - b).abs().max()
+    finally:
+        dist.destroy_process_group()
