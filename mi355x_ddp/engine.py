@@ -49,6 +49,21 @@ def _recover_failed_capture(ctx):
         torch._C._cuda_releasePool(dev, ctx.pool[0])
     except Exception:
         pass
+    try:
+        # capture_begin registers the default CUDA generator and flips its
+        # state to capture mode (capture_prologue); only a SUCCESSFUL
+        # capture_end flips it back. Nothing on the failure path clears
+        # it, so the next RNG op in the process would raise "Offset
+        # increment outside graph capture". Swap in a fresh state object
+        # carrying the same seed/offset (graphsafe_set_state shares the
+        # new state, whose capture flag starts clear).
+        dev = torch.cuda.current_device()
+        dg = torch.cuda.default_generators[dev]
+        fresh = torch.Generator(device=f"cuda:{dev}")
+        fresh.set_state(dg.get_state())
+        dg.graphsafe_set_state(fresh)
+    except Exception:
+        pass
     if ops.has_ext():
         ops.ext().clear_hip_errors()  # pop the latched HIP error
     torch.cuda.synchronize()

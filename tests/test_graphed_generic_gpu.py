@@ -325,6 +325,9 @@ def test_graphed_recovers_from_invalidated_capture():
     torch.cuda.synchronize()
     assert gs._broken
     assert any("capture failed" in str(x.message) for x in w)
+    # the default CUDA generator must be usable again (a failed capture
+    # leaves its capture flag stuck without the recovery's state swap)
+    assert torch.isfinite(torch.randn(8, device=DEV)).all()
 
     # eager arm: identical schedule (warmup trains, the poisoned capture
     # executed nothing, every step() ran eager)
