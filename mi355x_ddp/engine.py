@@ -416,6 +416,19 @@ class GraphedAutogradStep:
         self.flush()
         self._s_start, self._s_next = i, i + 1
 
+    def chunks(self, n):
+        """Greedy decomposition of an n-step run over chunk_sizes graphs
+        (largest-first; chunk_sizes always contains 1, so any n is
+        covered exactly). Pure logic — property-tested on CPU."""
+        out = []
+        while n:
+            for G in self.chunk_sizes:
+                if G <= n:
+                    break
+            out.append(G)
+            n -= G
+        return out
+
     def flush(self):
         if self._s_next <= self._s_start:
             return
@@ -425,10 +438,7 @@ class GraphedAutogradStep:
         if not self._warmed:
             self._warm(self._shard_x[lo * B:(lo + 1) * B],
                        self._shard_t[lo * B:(lo + 1) * B])
-        while n:
-            for G in self.chunk_sizes:
-                if G <= n:
-                    break
+        for G in self.chunks(n):
             s, e = lo * B, (lo + G) * B
             got = self._get_graph(G)
             if got is None:  # capture broken: eager the rest of the run

@@ -222,3 +222,28 @@ def test_bench_world2_gloo_driver_form():
     assert r["config"]["parallelism"] == "dp2"
     assert r["config"]["comm"] == "gloo-cpu"
     assert r["value"] > 0
+
+
+def test_graphed_chunk_decomposition_properties():
+    """GraphedAutogradStep.chunks: every run length decomposes exactly,
+    largest-first, with the minimal greedy tail (pure logic, CPU)."""
+    from hypothesis import given, strategies as st
+
+    from mi355x_ddp.engine import GraphedAutogradStep
+
+    eng = GraphedAutogradStep.__new__(GraphedAutogradStep)
+    eng.chunk_sizes = (64, 8, 1)
+
+    @given(st.integers(min_value=1, max_value=5000))
+    def check(n):
+        cs = eng.chunks(n)
+        assert sum(cs) == n                      # covers exactly
+        assert all(c in (64, 8, 1) for c in cs)  # only known graphs
+        assert cs == sorted(cs, reverse=True)    # largest-first
+        # greedy is optimal for divisible chunk ladders (64 = 8*8):
+        assert cs.count(8) < 8 and cs.count(1) < 8
+
+    check()
+
+    eng.chunk_sizes = (5, 1)  # non-divisible ladder still covers exactly
+    assert eng.chunks(13) == [5, 5, 1, 1, 1]
