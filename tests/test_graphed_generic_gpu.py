@@ -346,3 +346,80 @@ def test_graphed_recovers_from_invalidated_capture():
     torch.cuda.synchronize()
     for a, b in zip(m.parameters(), m_e.parameters()):
         assert torch.equal(a, b), (a - b).abs().max()
+
+
+def test_graphed_state_machine_fuzz():
+    """Randomized interleaving of the engine's surfaces — sequential
+    step_shard runs, run-breaking random indices, tensor-API step()
+    calls, explicit flushes, shard re-binds — against an eager oracle
+    that replays the exact submission order (mirror of the persistent
+    engine's fuzz, test_engine_gpu.py)."""
+    import random
+    rng = random.Random(1234)
+    S, B = 24, 16
+    g = torch.Generator().manual_seed(99)
+    xs = torch.randn(S * B, 64, generator=g).to(DEV)
+    ts = torch.randn(S * B, 10, generator=g).to(DEV)
+    pool_x = torch.randn(8 * B, 64, generator=g).to(DEV)
+    pool_t = torch.randn(8 * B, 10, generator=g).to(DEV)
+
+    for trial in range(3):
+        m, eng, opt = _build(100 + trial)
+        gs = GraphedAutogradStep(eng, ops.mse_loss, opt,
+                                 finalize=eng.finalize_backward,
+                                 chunk_sizes=(8, 1))
+        m_e, eng_e, opt_e = _build(100 + trial)
+
+        def eager(x, t):
+            loss = ops.mse_loss(eng_e(x), t)
+            loss.backward()
+            eng_e.finalize_backward()
+            opt_e.step()
+
+        submitted = []
+        gs.bind_shard(xs, ts, B)
+        next_seq = 0
+        warm_done = False
+        for op in range(40):
+            r = rng.random()
+            if r < 0.45:  # sequential shard index
+                i = next_seq % S
+                next_seq += 1
+                gs.step_shard(i)
+                submitted.append((xs[i * B:(i + 1) * B],
+                                  ts[i * B:(i + 1) * B]))
+            elif r < 0.65:  # random index (breaks the run)
+                i = rng.randrange(S)
+                next_seq = i + 1
+                gs.step_shard(i)
+                submitted.append((xs[i * B:(i + 1) * B],
+                                  ts[i * B:(i + 1) * B]))
+            elif r < 0.85:  # tensor-API step from a different buffer
+                j = rng.randrange(8)
+                x, t = (pool_x[j * B:(j + 1) * B],
+                        pool_t[j * B:(j + 1) * B])
+                gs.step(x, t)
+                submitted.append((x, t))
+            else:
+                gs.flush()
+            # the first executed work triggers the one-time warmup (3
+            # extra steps on that batch) — replicate in the oracle
+            if submitted and not warm_done:
+                # warmup happens when the first step/flush actually runs
+                if gs._warmed:
+                    warm_done = True
+                    x0, t0 = submitted[0]
+                    for _ in range(gs.warmup_steps):
+                        eager(x0, t0)
+        gs.flush()
+        torch.cuda.synchronize()
+        assert not gs._broken
+        if not warm_done and submitted and gs._warmed:
+            x0, t0 = submitted[0]
+            for _ in range(gs.warmup_steps):
+                eager(x0, t0)
+        for x, t in submitted:
+            eager(x, t)
+        torch.cuda.synchronize()
+        for a, b in zip(m.parameters(), m_e.parameters()):
+            assert torch.equal(a, b), (trial, (a - b).abs().max())
