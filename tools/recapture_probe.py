@@ -1,6 +1,8 @@
 """After a failed (invalidated) capture + recovery variant, can the
 process capture again? variant: both | noend | norelease | none"""
+import os
 import sys
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 variant = sys.argv[1]
