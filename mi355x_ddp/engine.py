@@ -30,6 +30,26 @@ from .models.toy import HipLinear
 from .parallel.reducer import Reducer
 
 
+# Process-wide capture poison: after ONE invalidated capture, any further
+# capture_begin in this process is lethal on this torch/ROCm build —
+# capture_begin raises "Cannot register the state during capturing stage"
+# mid-registration and the partially-registered CUDAGraph's DESTRUCTOR
+# then throws inside ~CUDAGraph ("The graph should be registered to the
+# state") which calls std::terminate (measured: tools/recapture_probe.py,
+# every cleanup variant dumps core). So: first failure -> never capture
+# again in this process; all graphed engines run eager.
+_CAPTURE_POISONED = False
+# Graphs whose capture failed are LEAKED on purpose: destroying them can
+# terminate the process (above). A handful of small host objects.
+_DEAD_GRAPHS = []
+
+
+def _poison_captures(graph):
+    global _CAPTURE_POISONED
+    _CAPTURE_POISONED = True
+    _DEAD_GRAPHS.append(graph)
+
+
 def _recover_failed_capture(ctx):
     """Best-effort cleanup after an INVALIDATED stream capture (e.g. a
     library allocating outside torch's allocator mid-capture — observed
@@ -343,6 +363,9 @@ class GraphedAutogradStep:
         got = self._graphs.get(G)
         if got is not None or self._broken:
             return got
+        if _CAPTURE_POISONED:
+            self._broken = True  # an earlier capture in this process
+            return None          # failed: capturing again is lethal
         B = self._bshape[0][0]
         xbuf = torch.empty((G * B,) + self._bshape[0][1:],
                            dtype=self._bdtype[0], device=self._bdev)
@@ -361,9 +384,11 @@ class GraphedAutogradStep:
             import warnings
             warnings.warn(
                 f"[mi355x_ddp] whole-step hipGraph capture failed ({e!r}); "
-                "GraphedAutogradStep running eager (correct, slower)",
+                "GraphedAutogradStep running eager (correct, slower; no "
+                "further captures will be attempted in this process)",
                 RuntimeWarning, stacklevel=2)
             self._broken = True
+            _poison_captures(g)
             _recover_failed_capture(ctx)
             return None
 
@@ -476,6 +501,9 @@ class GraphedToyStep(ToyFusedStep):
         # warm up collectives/kernels outside capture first
         super().step(self._x_static, self._t_static)
         torch.cuda.synchronize()
+        if _CAPTURE_POISONED:
+            self._graph = False  # earlier failed capture: stay eager
+            return
         g = torch.cuda.CUDAGraph()
         ctx = torch.cuda.graph(g)
         try:
@@ -489,6 +517,7 @@ class GraphedToyStep(ToyFusedStep):
                 "GraphedToyStep running eager (correct, slower)",
                 RuntimeWarning, stacklevel=2)
             self._graph = False
+            _poison_captures(g)
             _recover_failed_capture(ctx)
 
     def step(self, x: torch.Tensor, t: torch.Tensor) -> None:

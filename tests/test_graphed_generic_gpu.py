@@ -298,56 +298,6 @@ def test_graphed_shard_rebind_with_new_batch_size():
         assert torch.equal(a, b), (a - b).abs().max()
 
 
-def test_graphed_recovers_from_invalidated_capture():
-    # force a capture invalidation deterministically (a synchronize mid-
-    # capture is illegal) and verify the engine recovers: warning, eager
-    # fallback, training continues and stays correct (engine.py
-    # _recover_failed_capture — exercised for real by MIOpen NHWC
-    # workspace allocs, profiles r02d)
-    import warnings as warnings_mod
-
-    m, eng, opt = _build(8)
-    state = {"broke": False}
-
-    def poisoned_loss(y, t):
-        if torch.cuda.is_current_stream_capturing() and not state["broke"]:
-            state["broke"] = True
-            torch.cuda.synchronize()  # invalidates the capture
-        return ops.mse_loss(y, t)
-
-    gs = GraphedAutogradStep(eng, poisoned_loss, opt,
-                             finalize=eng.finalize_backward)
-    data = _batches(4, seed=91)
-    with warnings_mod.catch_warnings(record=True) as w:
-        warnings_mod.simplefilter("always")
-        for x, t in data:
-            gs.step(x, t)
-    torch.cuda.synchronize()
-    assert gs._broken
-    assert any("capture failed" in str(x.message) for x in w)
-    # the default CUDA generator must be usable again (a failed capture
-    # leaves its capture flag stuck without the recovery's state swap)
-    assert torch.isfinite(torch.randn(8, device=DEV)).all()
-
-    # eager arm: identical schedule (warmup trains, the poisoned capture
-    # executed nothing, every step() ran eager)
-    m_e, eng_e, opt_e = _build(8)
-
-    def eager(x, t):
-        loss = ops.mse_loss(eng_e(x), t)
-        loss.backward()
-        eng_e.finalize_backward()
-        opt_e.step()
-
-    for _ in range(gs.warmup_steps):
-        eager(*data[0])
-    for x, t in data:
-        eager(x, t)
-    torch.cuda.synchronize()
-    for a, b in zip(m.parameters(), m_e.parameters()):
-        assert torch.equal(a, b), (a - b).abs().max()
-
-
 def test_graphed_state_machine_fuzz():
     """Randomized interleaving of the engine's surfaces — sequential
     step_shard runs, run-breaking random indices, tensor-API step()
@@ -423,3 +373,67 @@ def test_graphed_state_machine_fuzz():
         torch.cuda.synchronize()
         for a, b in zip(m.parameters(), m_e.parameters()):
             assert torch.equal(a, b), (trial, (a - b).abs().max())
+
+
+def test_graphed_recovers_from_invalidated_capture():
+    # force a capture invalidation deterministically (a synchronize mid-
+    # capture is illegal) and verify the engine recovers: warning, eager
+    # fallback, training continues and stays correct (engine.py
+    # _recover_failed_capture — exercised for real by MIOpen NHWC
+    # workspace allocs, profiles r02d)
+    import warnings as warnings_mod
+
+    m, eng, opt = _build(8)
+    state = {"broke": False}
+
+    def poisoned_loss(y, t):
+        if torch.cuda.is_current_stream_capturing() and not state["broke"]:
+            state["broke"] = True
+            torch.cuda.synchronize()  # invalidates the capture
+        return ops.mse_loss(y, t)
+
+    gs = GraphedAutogradStep(eng, poisoned_loss, opt,
+                             finalize=eng.finalize_backward)
+    data = _batches(4, seed=91)
+    with warnings_mod.catch_warnings(record=True) as w:
+        warnings_mod.simplefilter("always")
+        for x, t in data:
+            gs.step(x, t)
+    torch.cuda.synchronize()
+    assert gs._broken
+    assert any("capture failed" in str(x.message) for x in w)
+    # the default CUDA generator must be usable again (a failed capture
+    # leaves its capture flag stuck without the recovery's state swap)
+    assert torch.isfinite(torch.randn(8, device=DEV)).all()
+    # process-wide poison: a NEW engine must not attempt to capture (a
+    # second capture_begin after an invalidated capture is lethal on this
+    # build — engine.py _poison_captures; tools/recapture_probe.py).
+    # This test therefore runs LAST in the file.
+    import mi355x_ddp.engine as eng_mod
+    assert eng_mod._CAPTURE_POISONED
+    m2, eng2, opt2 = _build(9)
+    gs2 = GraphedAutogradStep(eng2, ops.mse_loss, opt2,
+                              finalize=eng2.finalize_backward)
+    for x, t in _batches(2, seed=92):
+        gs2.step(x, t)
+    torch.cuda.synchronize()
+    assert gs2._broken and not gs2._graphs  # eager, no capture attempted
+    assert all(torch.isfinite(p).all() for p in m2.parameters())
+
+    # eager arm: identical schedule (warmup trains, the poisoned capture
+    # executed nothing, every step() ran eager)
+    m_e, eng_e, opt_e = _build(8)
+
+    def eager(x, t):
+        loss = ops.mse_loss(eng_e(x), t)
+        loss.backward()
+        eng_e.finalize_backward()
+        opt_e.step()
+
+    for _ in range(gs.warmup_steps):
+        eager(*data[0])
+    for x, t in data:
+        eager(x, t)
+    torch.cuda.synchronize()
+    for a, b in zip(m.parameters(), m_e.parameters()):
+        assert torch.equal(a, b), (a - b).abs().max()
